@@ -1,0 +1,146 @@
+"""Product host plumbing vs the oracle (itself pinned to the reference).
+
+The product's cell-list RGG must produce the BIT-IDENTICAL per-rank CSR the
+reference's O((nv/p)^2) generator produces (graph.hpp:584-1213), including
+the adjacent-rank equal-local-index quirk and -w Euclidean weights.
+"""
+import os
+import tempfile
+
+import numpy as np
+import pytest
+
+from minivite_amd import Graph
+from oracle.oracle import OracleGraph
+
+
+@pytest.mark.parametrize("nv,p,unit", [
+    (16384, 1, True),
+    (16384, 2, True),
+    (16384, 4, True),
+    (16384, 8, True),
+    (16384, 2, False),
+])
+def test_rgg_bit_identical_to_oracle(nv, p, unit):
+    og = OracleGraph.rgg(nv, p, unit_weight=unit)
+    try:
+        for r in range(p):
+            pg = Graph.rgg(nv, rank=r, nranks=p, unit_weight=unit)
+            try:
+                oxa, ota, owa = og.rank_arrays(r)
+                xa, ta, wa = pg.arrays()
+                assert np.array_equal(xa, oxa)
+                assert np.array_equal(ta, ota)
+                assert np.array_equal(wa, owa)
+            finally:
+                pg.free()
+    finally:
+        og.free()
+
+
+def test_binary_roundtrip():
+    """write_binary -> read_binary reproduces the graph (any nranks)."""
+    g = Graph.rgg(16384, 0, 1)
+    try:
+        with tempfile.TemporaryDirectory() as d:
+            path = os.path.join(d, "g.bin")
+            g.write_binary(path)
+            # whole-graph read at p=1
+            g1 = Graph.read_binary(path, 0, 1)
+            a, b, c = g.arrays()
+            x, y, z = g1.arrays()
+            assert np.array_equal(a, x) and np.array_equal(b, y) \
+                and np.array_equal(c, z)
+            g1.free()
+            # partitioned read at p=4: slices concatenate back to the whole
+            xs, ts, ws = [], [], []
+            off = 0
+            for r in range(4):
+                gr = Graph.read_binary(path, r, 4)
+                xa, ta, wa = gr.arrays()
+                xs.append(xa[:-1] + off if r < 3 else xa + off)
+                off += xa[-1]
+                ts.append(ta)
+                ws.append(wa)
+                gr.free()
+            assert np.array_equal(np.concatenate(xs), a)
+            assert np.array_equal(np.concatenate(ts), b)
+            assert np.array_equal(np.concatenate(ws), c)
+    finally:
+        g.free()
+
+
+def test_balanced_read_partition():
+    """-b balanced read: per-rank edges within 2x of the mean, all vertices
+    covered (find_balanced_num_edges semantics, graph.hpp:416-461)."""
+    g = Graph.rgg(16384, 0, 1)
+    try:
+        with tempfile.TemporaryDirectory() as d:
+            path = os.path.join(d, "g.bin")
+            g.write_binary(path)
+            total_lnv, total_lne = 0, 0
+            for r in range(4):
+                gr = Graph.read_binary(path, r, 4, balanced=True)
+                total_lnv += gr.lnv
+                total_lne += gr.lne
+                gr.free()
+            assert total_lnv == 16384
+            assert total_lne == g.lne
+    finally:
+        g.free()
+
+
+def test_oracle_consumes_product_graph():
+    """Louvain over a product-built (from_csr path) graph matches the pin —
+    the same-graph guarantee behind every CPU-baseline comparison."""
+    import json
+    pins = json.load(open(os.path.join(os.path.dirname(__file__), "golden",
+                                       "pins.json")))
+    pin = pins["rgg_n16384_p4_unit"]
+    from oracle.oracle import louvain
+    csrs = []
+    for r in range(4):
+        pg = Graph.rgg(16384, rank=r, nranks=4)
+        csrs.append(pg.arrays())
+        pg.free()
+    parts = np.array([(16384 * r) // 4 for r in range(5)], dtype=np.int64)
+    og = OracleGraph.from_csr(16384, 4, parts, [(x, t, w) for x, t, w in csrs])
+    try:
+        mod, iters = louvain(og)
+        assert iters == pin["iters"]
+        assert float(mod).hex() == pin["final_mod_hex"]
+    finally:
+        og.free()
+
+
+def test_rgg_random_edges_added():
+    """-p adds extra directed edges (perf-only semantics)."""
+    g0 = Graph.rgg(16384, 0, 1)
+    g1 = Graph.rgg(16384, 0, 1, random_edge_percent=4.0)
+    try:
+        assert g1.lne > g0.lne
+        # roughly 2*4% of undirected count = 4% of directed
+        extra = g1.lne - g0.lne
+        assert 0.5 * 0.04 * g0.lne < extra < 2.5 * 0.04 * g0.lne
+    finally:
+        g0.free()
+        g1.free()
+
+
+def test_rgg_p_reciprocal_edges_consistent():
+    """With -p at nranks>1, rank A's extra edge to B appears on B too."""
+    gs = [Graph.rgg(16384, r, 2, random_edge_percent=2.0) for r in range(2)]
+    try:
+        edges = set()
+        for r, g in enumerate(gs):
+            xa, ta, _ = g.arrays()
+            base = (16384 * r) // 2
+            for i in range(g.lnv):
+                for e in range(xa[i], xa[i + 1]):
+                    edges.add((base + i, ta[e]))
+        # every directed edge's reverse exists
+        missing = [(u, v) for (u, v) in edges if (v, u) not in edges]
+        assert not missing, f"{len(missing)} unreciprocated edges"
+    finally:
+        for g in gs:
+            g.free()
