@@ -71,6 +71,10 @@ const int64_t *mv_graph_parts(const mv_graph *g); /* nranks+1 */
 const int64_t *mv_graph_xadj(const mv_graph *g);  /* lnv+1 */
 const int64_t *mv_graph_tails(const mv_graph *g); /* lne */
 const double *mv_graph_weights(const mv_graph *g);/* lne */
+/* Internal-layout hint (NULL if absent): locality_perm[k] = original local
+ * vertex id at internal position k, spatially ordered. Layout metadata
+ * only — results are identical with or without it. */
+const int32_t *mv_graph_locality_hint(const mv_graph *g);
 
 /* ---- the hot path (replaces distLouvainMethod, dspl.hpp:1280-1441) ---- */
 

@@ -10,11 +10,33 @@
 // sites become grouped ncclSend/ncclRecv alltoallv over
 // xGMI + ncclAllReduce for the 1-2 double reductions.
 //
-// FP discipline: built with -ffp-contract=off so the ΔQ gain expression
+// Device data layout:
+//  * Per-vertex working arrays (currComm/pastComm/targetComm, vDegree,
+//    clusterWeight) live in INTERNAL order: a spatial permutation sigma
+//    (the builder's locality hint) so that a vertex's neighbors — whose
+//    GLOBAL ids the RGG assigns randomly — sit nearby in memory and the
+//    per-edge community gathers hit L1/L2 instead of round-tripping to the
+//    Infinity Cache (measured: the label-order layout fetched 3.5x its
+//    algorithmic bytes at 13% L2 hit rate, 86% wave-wait). Community
+//    LABELS, the wire protocol and every result are untouched: sigma is
+//    pure layout. Without a hint, internal order = degree-sorted (load
+//    balance for skewed graphs).
+//  * Edges: SELL-64 (sliced ELL, slice = one wave64) over internal order;
+//    edge step k of a wave loads 64 consecutive int32 pre-translated tails
+//    (local -> internal index, ghost -> lnv + slot in the sorted ghost
+//    list) — one coalesced 256-B line pair per wave-instruction, replacing
+//    the reference's 16-B AoS row walk (graph.hpp:60-66) and per-edge hash
+//    lookup (dspl.hpp:253-260). Unit-weight graphs skip the weight stream
+//    entirely (every w == 1.0, detected at load).
+//  * Community info (localCinfo/localCupdate, dspl.hpp:61-66) is AoS
+//    {int64 size; double degree} indexed by community label - base, so one
+//    lookup costs one cache line, not two.
+//
+// FP discipline: built with -ffp-contract=off so the dQ gain expression
 // (dspl.hpp:212) and all accumulations carry the same bits as the
 // gcc-built reference / oracle (generic x86-64 has no FMA contraction).
 // Per-vertex weight accumulation is sequential in edge order (one lane owns
-// one vertex), matching dspl.hpp:240-271 exactly, so -w results align too.
+// one vertex), matching dspl.hpp:240-271 exactly.
 //
 // This is GPU-only code: engine creation fails loudly without a device.
 // There is no CPU fallback anywhere in this file.
@@ -25,6 +47,7 @@
 #include <cstdio>
 #include <cstdlib>
 #include <cstring>
+#include <type_traits>
 #include <vector>
 
 #include <hip/hip_runtime.h>
@@ -57,9 +80,9 @@ namespace {
 
 using i64 = int64_t;
 
-struct DevEdge {   // device edge record, 16 B (mirrors Edge, graph.hpp:60-66,
-    i64 tidx;      // with the tail pre-translated: local i -> i, ghost ->
-    double w;      // lnv + slot in the sorted ghost list)
+struct Cinfo {     // localCinfo / localCupdate entry (Comm, dspl.hpp:61-66)
+    i64 size;
+    double degree;
 };
 
 struct Info16 {    // wire record for cinfo replies / delta routing: the
@@ -87,26 +110,31 @@ __device__ __forceinline__ void atomic_add_i64(i64 *p, i64 v) {
               static_cast<unsigned long long>(v));
 }
 
-// ---- K1: vDegree + localCinfo init (dspl.hpp:82-107) ----
-__global__ void k1_vertex_degree(i64 lnv, const i64 *__restrict__ xadj,
-                                 const double *__restrict__ ew,
+// ---- K1: vDegree + localCinfo init (dspl.hpp:82-107). vDegree is
+// internal-ordered; cinfo is label-ordered. ----
+__global__ void k1_vertex_degree(i64 lnv, const unsigned *__restrict__ sigma,
+                                 const i64 *__restrict__ xadj,
+                                 const double *__restrict__ ew, int unit,
                                  double *__restrict__ vDegree,
-                                 i64 *__restrict__ cinfo_size,
-                                 double *__restrict__ cinfo_degree) {
-    for (i64 i = blockIdx.x * (i64)blockDim.x + threadIdx.x; i < lnv;
-         i += (i64)gridDim.x * blockDim.x) {
-        double tw = 0.0;
-        const i64 e1 = xadj[i + 1];
-        for (i64 e = xadj[i]; e < e1; e++) tw += ew[e]; // sequential edge order
-        vDegree[i] = tw;
-        cinfo_degree[i] = tw;
-        cinfo_size[i] = 1; // dspl.hpp:104-105
+                                 Cinfo *__restrict__ cinfo) {
+    for (i64 k = blockIdx.x * (i64)blockDim.x + threadIdx.x; k < lnv;
+         k += (i64)gridDim.x * blockDim.x) {
+        const i64 v = sigma[k];
+        const i64 e0 = xadj[v], e1 = xadj[v + 1];
+        double tw;
+        if (unit) {
+            tw = (double)(e1 - e0); // unit weights: exact
+        } else {
+            tw = 0.0;
+            for (i64 e = e0; e < e1; e++) tw += ew[e]; // sequential edge order
+        }
+        vDegree[k] = tw;
+        cinfo[v] = {1, tw}; // dspl.hpp:104-105
     }
 }
 
-// block-partial sum for K2/K7 (deterministic: fixed block ranges, lane-
-// strided partials reduced by a fixed shuffle tree; partials summed on host
-// in block order)
+// block-partial sum for K2/K7 (deterministic: fixed block ranges, fixed
+// shuffle tree; partials summed on host in block order)
 template <typename F>
 __global__ void k_partial_sum2(i64 n, F f, double *__restrict__ out2) {
     double a = 0.0, b = 0.0;
@@ -134,13 +162,16 @@ __global__ void k_partial_sum2(i64 n, F f, double *__restrict__ out2) {
     }
 }
 
-// ---- K3: community iota (dspl.hpp:132-149) ----
-__global__ void k3_init_comm(i64 lnv, i64 base, i64 *__restrict__ curr,
-                             i64 *__restrict__ past) {
-    for (i64 i = blockIdx.x * (i64)blockDim.x + threadIdx.x; i < lnv;
-         i += (i64)gridDim.x * blockDim.x) {
-        curr[i] = i + base;
-        past[i] = i + base;
+// ---- K3: community iota (dspl.hpp:132-149): internal slot k holds the
+// LABEL of the vertex it represents ----
+__global__ void k3_init_comm(i64 lnv, i64 base,
+                             const unsigned *__restrict__ sigma,
+                             i64 *__restrict__ curr, i64 *__restrict__ past) {
+    for (i64 k = blockIdx.x * (i64)blockDim.x + threadIdx.x; k < lnv;
+         k += (i64)gridDim.x * blockDim.x) {
+        const i64 label = (i64)sigma[k] + base;
+        curr[k] = label;
+        past[k] = label;
     }
 }
 
@@ -155,34 +186,94 @@ __global__ void k_select_remote(i64 lne, const i64 *__restrict__ tails,
     }
 }
 
-// ---- K10b: translate tails into DevEdge (product layout; removes the
-// reference's per-edge unordered_map lookup, dspl.hpp:253-260) ----
-__global__ void k_build_edges(i64 lne, const i64 *__restrict__ tails,
-                              const double *__restrict__ w, i64 base,
-                              i64 bound, i64 lnv,
-                              const i64 *__restrict__ ghosts, i64 nghost,
-                              DevEdge *__restrict__ edges) {
-    for (i64 e = blockIdx.x * (i64)blockDim.x + threadIdx.x; e < lne;
-         e += (i64)gridDim.x * blockDim.x) {
-        const i64 t = tails[e];
-        i64 ti = (t >= base && t < bound)
-                     ? t - base
-                     : lnv + dev_lower_bound(ghosts, nghost, t);
-        edges[e] = {ti, w[e]};
+// ---- SELL build (per run, setup span) ----
+__global__ void k_degrees(i64 lnv, const unsigned *__restrict__ sigma,
+                          const i64 *__restrict__ xadj,
+                          unsigned *__restrict__ deg) {
+    for (i64 k = blockIdx.x * (i64)blockDim.x + threadIdx.x; k < lnv;
+         k += (i64)gridDim.x * blockDim.x) {
+        const i64 v = sigma[k];
+        deg[k] = (unsigned)(xadj[v + 1] - xadj[v]);
     }
 }
 
-// ---- K8: scdata gather (dspl.hpp:559-571): comm of each exported vertex ----
-__global__ void k8_gather_comms(i64 n, const i64 *__restrict__ svdata,
-                                i64 base, const i64 *__restrict__ currComm,
+__global__ void k_iota32(i64 n, unsigned *__restrict__ out) {
+    for (i64 i = blockIdx.x * (i64)blockDim.x + threadIdx.x; i < n;
+         i += (i64)gridDim.x * blockDim.x)
+        out[i] = (unsigned)i;
+}
+
+__global__ void k_invert_perm(i64 n, const unsigned *__restrict__ perm,
+                              unsigned *__restrict__ inv) {
+    for (i64 i = blockIdx.x * (i64)blockDim.x + threadIdx.x; i < n;
+         i += (i64)gridDim.x * blockDim.x)
+        inv[perm[i]] = (unsigned)i;
+}
+
+// slice widths: chunk c spans 64 SELL positions, width = its max degree;
+// emitted as element counts (width*64)
+__global__ void k_chunk_sizes(i64 nchunks, i64 lnv,
+                              const unsigned *__restrict__ perm,
+                              const unsigned *__restrict__ deg,
+                              i64 *__restrict__ sizes) {
+    for (i64 c = blockIdx.x * (i64)blockDim.x + threadIdx.x; c < nchunks;
+         c += (i64)gridDim.x * blockDim.x) {
+        unsigned w = 0;
+        const i64 s1 = min(c * 64 + 64, lnv);
+        for (i64 s = c * 64; s < s1; s++) w = max(w, deg[perm[s]]);
+        sizes[c] = (i64)w * 64;
+    }
+}
+
+// translate + scatter this rank's CSR rows into the SELL image
+__global__ void k_fill_sell(i64 lnv, const unsigned *__restrict__ perm,
+                            const unsigned *__restrict__ sigma,
+                            const unsigned *__restrict__ sigma_inv,
+                            const i64 *__restrict__ xadj,
+                            const i64 *__restrict__ tails,
+                            const double *__restrict__ w, i64 base, i64 bound,
+                            const i64 *__restrict__ ghosts, i64 nghost,
+                            const i64 *__restrict__ chunk_off,
+                            int *__restrict__ sell_tidx,
+                            double *__restrict__ sell_w) {
+    for (i64 s = blockIdx.x * (i64)blockDim.x + threadIdx.x; s < lnv;
+         s += (i64)gridDim.x * blockDim.x) {
+        const i64 iint = perm[s];
+        const i64 v = sigma[iint];
+        const i64 off = chunk_off[s >> 6];
+        const int l = (int)(s & 63);
+        const i64 e0 = xadj[v], e1 = xadj[v + 1];
+        for (i64 e = e0; e < e1; e++) {
+            const i64 t = tails[e];
+            const i64 ti = (t >= base && t < bound)
+                               ? (i64)sigma_inv[t - base]
+                               : lnv + dev_lower_bound(ghosts, nghost, t);
+            const i64 slot = off + (e - e0) * 64 + l;
+            sell_tidx[slot] = (int)ti;
+            if (sell_w) sell_w[slot] = w[e];
+        }
+    }
+}
+
+// translate a global-id list into internal indices (svdata, once per run)
+__global__ void k_to_internal(i64 n, const i64 *__restrict__ gids, i64 base,
+                              const unsigned *__restrict__ sigma_inv,
+                              unsigned *__restrict__ out) {
+    for (i64 k = blockIdx.x * (i64)blockDim.x + threadIdx.x; k < n;
+         k += (i64)gridDim.x * blockDim.x)
+        out[k] = sigma_inv[gids[k] - base];
+}
+
+// ---- K8: scdata gather (dspl.hpp:559-571) ----
+__global__ void k8_gather_comms(i64 n, const unsigned *__restrict__ svdata_int,
+                                const i64 *__restrict__ currComm,
                                 i64 *__restrict__ out) {
     for (i64 k = blockIdx.x * (i64)blockDim.x + threadIdx.x; k < n;
          k += (i64)gridDim.x * blockDim.x)
-        out[k] = currComm[svdata[k] - base];
+        out[k] = currComm[svdata_int[k]];
 }
 
-// ---- candidate remote communities (dspl.hpp:670-700): ghost comms +
-// own currComm, filtered to remote owners ----
+// ---- candidate remote communities (dspl.hpp:670-700) ----
 __global__ void k_filter_remote(i64 n, const i64 *__restrict__ vals, i64 base,
                                 i64 bound, i64 *__restrict__ out,
                                 unsigned long long *__restrict__ count) {
@@ -203,84 +294,68 @@ __global__ void k_owner_bounds(const i64 *__restrict__ sorted, i64 n,
 
 // ---- K9: cinfo reply gather (dspl.hpp:776-929) ----
 __global__ void k9_reply_info(i64 n, const i64 *__restrict__ req_ids, i64 base,
-                              const i64 *__restrict__ cinfo_size,
-                              const double *__restrict__ cinfo_degree,
+                              const Cinfo *__restrict__ cinfo,
                               Info16 *__restrict__ out) {
     for (i64 k = blockIdx.x * (i64)blockDim.x + threadIdx.x; k < n;
          k += (i64)gridDim.x * blockDim.x) {
-        const i64 c = req_ids[k] - base;
-        out[k] = {cinfo_size[c], cinfo_degree[c]};
+        const Cinfo c = cinfo[req_ids[k] - base];
+        out[k] = {c.size, c.degree};
     }
-}
-
-// unpack received rc_info into SoA
-__global__ void k_unpack_info(i64 n, const Info16 *__restrict__ in,
-                              i64 *__restrict__ size_out,
-                              double *__restrict__ deg_out) {
-    for (i64 k = blockIdx.x * (i64)blockDim.x + threadIdx.x; k < n;
-         k += (i64)gridDim.x * blockDim.x) {
-        size_out[k] = in[k].size;
-        deg_out[k] = in[k].degree;
-    }
-}
-
-// pack remote-community deltas for halo #2 (updateRemoteCommunities,
-// dspl.hpp:988-1004: every remoteCinfo key is sent, zeros included)
-__global__ void k_pack_deltas(i64 n, const i64 *__restrict__ rcu_size,
-                              const double *__restrict__ rcu_degree,
-                              Info16 *__restrict__ out) {
-    for (i64 k = blockIdx.x * (i64)blockDim.x + threadIdx.x; k < n;
-         k += (i64)gridDim.x * blockDim.x)
-        out[k] = {rcu_size[k], rcu_degree[k]};
 }
 
 // apply received deltas to owned cinfo (dspl.hpp:1089-1102; atomics because
 // several senders may address one community)
 __global__ void k_apply_deltas(i64 n, const i64 *__restrict__ ids, i64 base,
                                const Info16 *__restrict__ deltas,
-                               i64 *__restrict__ cinfo_size,
-                               double *__restrict__ cinfo_degree) {
+                               Cinfo *__restrict__ cinfo) {
     for (i64 k = blockIdx.x * (i64)blockDim.x + threadIdx.x; k < n;
          k += (i64)gridDim.x * blockDim.x) {
-        const i64 c = ids[k] - base;
-        atomic_add_i64(&cinfo_size[c], deltas[k].size);
-        atomicAdd(&cinfo_degree[c], deltas[k].degree);
+        Cinfo *c = &cinfo[ids[k] - base];
+        atomic_add_i64(&c->size, deltas[k].size);
+        atomicAdd(&c->degree, deltas[k].degree);
     }
 }
 
 // ---- K6: apply localCupdate (dspl.hpp:458-471) ----
-__global__ void k6_apply_local(i64 lnv, const i64 *__restrict__ cupd_size,
-                               const double *__restrict__ cupd_degree,
-                               i64 *__restrict__ cinfo_size,
-                               double *__restrict__ cinfo_degree) {
+__global__ void k6_apply_local(i64 lnv, const Cinfo *__restrict__ cupd,
+                               Cinfo *__restrict__ cinfo) {
     for (i64 i = blockIdx.x * (i64)blockDim.x + threadIdx.x; i < lnv;
          i += (i64)gridDim.x * blockDim.x) {
-        cinfo_size[i] += cupd_size[i];
-        cinfo_degree[i] += cupd_degree[i];
+        cinfo[i].size += cupd[i].size;
+        cinfo[i].degree += cupd[i].degree;
     }
 }
 
+// de-permute an internal-ordered array into label order (trace only)
+__global__ void k_depermute(i64 lnv, const unsigned *__restrict__ sigma_inv,
+                            const i64 *__restrict__ in,
+                            i64 *__restrict__ out) {
+    for (i64 v = blockIdx.x * (i64)blockDim.x + threadIdx.x; v < lnv;
+         v += (i64)gridDim.x * blockDim.x)
+        out[v] = in[sigma_inv[v]];
+}
+
 // ---- K4: THE sweep (distExecuteLouvainIteration, dspl.hpp:276-405) ----
-// One lane per vertex; the clmap/counter hash (dspl.hpp:230-274) lives as
-// per-lane slot arrays: the first SLOTS distinct neighbor communities in
-// LDS (lane-strided, conflict-free for ds b64), the tail in a per-thread
-// global spill region (rare after iteration 1; L2-resident). The current
-// community's accumulator is held in a register (the reference's
-// counter[0], dspl.hpp:312-318). Per-lane sequential edge walk keeps -w
-// accumulation in edge order.
-template <int SLOTS>
+// One lane per vertex over the SELL image (see the layout note at the top).
+// The clmap/counter hash (dspl.hpp:230-274) lives as per-lane slot arrays:
+// the first SLOTS distinct neighbor communities in LDS (lane-strided,
+// conflict-free for ds b64), the tail in a per-thread global spill region
+// (only populated while communities are still fine-grained; L2-resident).
+// The current community's accumulator is a register (the reference's
+// counter[0], dspl.hpp:312-318).
+template <int SLOTS, bool UNIT>
 __global__ __launch_bounds__(256) void k4_sweep(
-    i64 lnv, i64 base, i64 bound, const i64 *__restrict__ xadj,
-    const DevEdge *__restrict__ edges, const i64 *__restrict__ currComm,
-    const i64 *__restrict__ ghost_comm, const double *__restrict__ vDegree,
-    const i64 *__restrict__ cinfo_size, const double *__restrict__ cinfo_degree,
-    i64 *__restrict__ cupd_size, double *__restrict__ cupd_degree,
-    const i64 *__restrict__ rc_ids, i64 nrc, const i64 *__restrict__ rc_size,
-    const double *__restrict__ rc_degree, i64 *__restrict__ rcu_size,
-    double *__restrict__ rcu_degree, double constant,
-    i64 *__restrict__ targetComm, double *__restrict__ clusterWeight,
-    i64 *__restrict__ spill_keys, double *__restrict__ spill_acc,
-    int spill_max) {
+    i64 lnv, i64 base, i64 bound, const unsigned *__restrict__ perm,
+    const unsigned *__restrict__ deg_int, const i64 *__restrict__ chunk_off,
+    const int *__restrict__ sell_tidx, const double *__restrict__ sell_w,
+    const i64 *__restrict__ currComm, const i64 *__restrict__ ghost_comm,
+    const double *__restrict__ vDegree, const unsigned *__restrict__ sigma,
+    const Cinfo *__restrict__ cinfo, Cinfo *__restrict__ cupd,
+    const i64 *__restrict__ rc_ids, i64 nrc,
+    const Info16 *__restrict__ rc_info, Info16 *__restrict__ rcu,
+    double constant, i64 *__restrict__ targetComm,
+    double *__restrict__ clusterWeight, i64 *__restrict__ spill_keys,
+    double *__restrict__ spill_acc, int spill_max) {
     extern __shared__ __attribute__((aligned(16))) char smem[];
     i64 *skey = reinterpret_cast<i64 *>(smem);
     double *sacc = reinterpret_cast<double *>(smem + sizeof(i64) * SLOTS * blockDim.x);
@@ -290,35 +365,39 @@ __global__ __launch_bounds__(256) void k4_sweep(
     i64 *myspill_k = spill_keys + gthread * spill_max;
     double *myspill_a = spill_acc + gthread * spill_max;
 
-    for (i64 i = gthread; i < lnv; i += stride) {
+    for (i64 s = gthread; s < lnv; s += stride) {
+        const i64 i = perm[s];          // internal vertex index
+        const int deg = (int)deg_int[i];
+        const i64 ebase = chunk_off[s >> 6] + (s & 63);
         const i64 cc = currComm[i];
         double ccDeg;
         i64 ccSize;
         if (cc >= base && cc < bound) { // dspl.hpp:296-307
-            ccDeg = cinfo_degree[cc - base];
-            ccSize = cinfo_size[cc - base];
+            const Cinfo c = cinfo[cc - base];
+            ccDeg = c.degree;
+            ccSize = c.size;
         } else {
-            const i64 s = dev_bsearch(rc_ids, nrc, cc);
-            ccDeg = rc_degree[s];
-            ccSize = rc_size[s];
+            const i64 q = dev_bsearch(rc_ids, nrc, cc);
+            ccDeg = rc_info[q].degree;
+            ccSize = rc_info[q].size;
         }
-        const i64 e0 = xadj[i], e1 = xadj[i + 1];
         i64 target;
-        if (e0 == e1) {
+        if (deg == 0) {
             target = cc; // dspl.hpp:323-324
         } else {
             double c0 = 0.0, selfLoop = 0.0;
             int ns = 0, nspill = 0;
-            for (i64 e = e0; e < e1; e++) {
-                const DevEdge ed = edges[e];
-                if (ed.tidx == i) selfLoop += ed.w; // dspl.hpp:247-248
-                const i64 tcomm = (ed.tidx < lnv) ? currComm[ed.tidx]
-                                                  : ghost_comm[ed.tidx - lnv];
-                if (tcomm == cc) { c0 += ed.w; continue; }
+            for (int k = 0; k < deg; k++) {
+                const i64 tidx = sell_tidx[ebase + (i64)k * 64];
+                const double w = UNIT ? 1.0 : sell_w[ebase + (i64)k * 64];
+                if (tidx == i) selfLoop += w; // dspl.hpp:247-248
+                const i64 tcomm = (tidx < lnv) ? currComm[tidx]
+                                               : ghost_comm[tidx - lnv];
+                if (tcomm == cc) { c0 += w; continue; }
                 bool found = false;
-                for (int s = 0; s < ns; s++) {
-                    if (skey[s * blockDim.x + tid] == tcomm) {
-                        sacc[s * blockDim.x + tid] += ed.w;
+                for (int t = 0; t < ns; t++) {
+                    if (skey[t * blockDim.x + tid] == tcomm) {
+                        sacc[t * blockDim.x + tid] += w;
                         found = true;
                         break;
                     }
@@ -326,20 +405,20 @@ __global__ __launch_bounds__(256) void k4_sweep(
                 if (found) continue;
                 if (ns < SLOTS) {
                     skey[ns * blockDim.x + tid] = tcomm;
-                    sacc[ns * blockDim.x + tid] = ed.w;
+                    sacc[ns * blockDim.x + tid] = w;
                     ns++;
                     continue;
                 }
-                for (int s = 0; s < nspill; s++) {
-                    if (myspill_k[s] == tcomm) {
-                        myspill_a[s] += ed.w;
+                for (int t = 0; t < nspill; t++) {
+                    if (myspill_k[t] == tcomm) {
+                        myspill_a[t] += w;
                         found = true;
                         break;
                     }
                 }
-                if (!found) { // spill_max sized to the max degree: cannot overflow
+                if (!found) { // spill_max covers the max degree: no overflow
                     myspill_k[nspill] = tcomm;
-                    myspill_a[nspill] = ed.w;
+                    myspill_a[nspill] = w;
                     nspill++;
                 }
             }
@@ -351,20 +430,21 @@ __global__ __launch_bounds__(256) void k4_sweep(
             const double ax = ccDeg - vdeg;
             double maxGain = 0.0;
             i64 maxIndex = cc, maxSize = ccSize;
-            for (int s = 0; s < ns + nspill; s++) {
-                const i64 y = (s < ns) ? skey[s * blockDim.x + tid]
-                                       : myspill_k[s - ns];
-                const double eiy = (s < ns) ? sacc[s * blockDim.x + tid]
-                                            : myspill_a[s - ns];
+            for (int t = 0; t < ns + nspill; t++) {
+                const i64 y = (t < ns) ? skey[t * blockDim.x + tid]
+                                       : myspill_k[t - ns];
+                const double eiy = (t < ns) ? sacc[t * blockDim.x + tid]
+                                            : myspill_a[t - ns];
                 double ay;
                 i64 ysz;
                 if (y >= base && y < bound) {
-                    ay = cinfo_degree[y - base];
-                    ysz = cinfo_size[y - base];
+                    const Cinfo c = cinfo[y - base];
+                    ay = c.degree;
+                    ysz = c.size;
                 } else {
                     const i64 q = dev_bsearch(rc_ids, nrc, y);
-                    ay = rc_degree[q];
-                    ysz = rc_size[q];
+                    ay = rc_info[q].degree;
+                    ysz = rc_info[q].size;
                 }
                 const double curGain =
                     2.0 * (eiy - eix) - 2.0 * vdeg * (ay - ax) * constant; // :212
@@ -383,20 +463,22 @@ __global__ __launch_bounds__(256) void k4_sweep(
         if (target != cc) { // 4-case updates (dspl.hpp:331-399)
             const double vdeg = vDegree[i];
             if (cc >= base && cc < bound) {
-                atomicAdd(&cupd_degree[cc - base], -vdeg);
-                atomic_add_i64(&cupd_size[cc - base], -1);
+                Cinfo *u = &cupd[cc - base];
+                atomicAdd(&u->degree, -vdeg);
+                atomic_add_i64(&u->size, -1);
             } else {
-                const i64 s = dev_bsearch(rc_ids, nrc, cc);
-                atomicAdd(&rcu_degree[s], -vdeg);
-                atomic_add_i64(&rcu_size[s], -1);
+                const i64 q = dev_bsearch(rc_ids, nrc, cc);
+                atomicAdd(&rcu[q].degree, -vdeg);
+                atomic_add_i64(&rcu[q].size, -1);
             }
             if (target >= base && target < bound) {
-                atomicAdd(&cupd_degree[target - base], vdeg);
-                atomic_add_i64(&cupd_size[target - base], 1);
+                Cinfo *u = &cupd[target - base];
+                atomicAdd(&u->degree, vdeg);
+                atomic_add_i64(&u->size, 1);
             } else {
-                const i64 s = dev_bsearch(rc_ids, nrc, target);
-                atomicAdd(&rcu_degree[s], vdeg);
-                atomic_add_i64(&rcu_size[s], 1);
+                const i64 q = dev_bsearch(rc_ids, nrc, target);
+                atomicAdd(&rcu[q].degree, vdeg);
+                atomic_add_i64(&rcu[q].size, 1);
             }
         }
         targetComm[i] = target; // dspl.hpp:404
@@ -422,16 +504,27 @@ struct mv_engine {
     i64 *d_parts = nullptr;
     i64 *d_xadj = nullptr;
     i64 *d_tails = nullptr;   // raw global tails (kept for setup)
-    double *d_ew = nullptr;   // edge weights
-    DevEdge *d_edges = nullptr;
+    double *d_ew = nullptr;   // edge weights (CSR order)
     int unit_weights = 1;
     i64 max_degree = 0;
+
+    // internal layout
+    unsigned *d_sigma = nullptr;     // internal -> original local id
+    unsigned *d_sigma_inv = nullptr; // original local id -> internal
+    int has_hint = 0;
+
+    // SELL-64 image (built in the per-run setup span)
+    unsigned *d_deg = nullptr;       // degree per INTERNAL index
+    unsigned *d_iota = nullptr, *d_perm = nullptr; // SELL pos -> internal
+    i64 *d_chunk_off = nullptr;      // nchunks+1 element offsets
+    i64 nchunks = 0, sell_elems = 0;
+    int *d_sell_tidx = nullptr;
+    double *d_sell_w = nullptr;
 
     // per-run state (device)
     i64 *d_curr = nullptr, *d_past = nullptr, *d_target = nullptr;
     double *d_vdeg = nullptr, *d_cw = nullptr;
-    i64 *d_cinfo_size = nullptr, *d_cupd_size = nullptr;
-    double *d_cinfo_deg = nullptr, *d_cupd_deg = nullptr;
+    Cinfo *d_cinfo = nullptr, *d_cupd = nullptr;
     double *d_partials = nullptr; // 2 * nblocks
     double *d_red = nullptr;      // 2 doubles for allreduce
 
@@ -440,6 +533,7 @@ struct mv_engine {
     i64 nghost = 0;
     i64 *d_ghost_comm = nullptr; // per-iteration communities of ghosts
     i64 *d_svdata = nullptr;     // vertices peers want from me (global ids)
+    unsigned *d_svdata_int = nullptr;
     i64 ssz = 0;
     std::vector<i64> send_off, recv_off; // per-peer offsets into svdata / ghosts
     i64 *d_scdata = nullptr;             // packed comms to export
@@ -447,15 +541,13 @@ struct mv_engine {
     // remote community info (per iteration)
     i64 rc_cap = 0;
     i64 *d_cand = nullptr, *d_cand_sorted = nullptr;
-    i64 *d_rc_ids = nullptr, *d_rc_size = nullptr;
-    double *d_rc_degree = nullptr;
-    i64 *d_rcu_size = nullptr;
-    double *d_rcu_degree = nullptr;
-    Info16 *d_rc_info = nullptr;
+    i64 *d_rc_ids = nullptr;
+    Info16 *d_rc_info = nullptr; // aligned with rc_ids
+    Info16 *d_rcu = nullptr;     // remoteCupdate accumulators
     i64 req_cap = 0;
-    i64 *d_req_ids = nullptr;   // ids other ranks requested from me
+    i64 *d_req_ids = nullptr;    // ids other ranks requested from me
     Info16 *d_req_info = nullptr;
-    i64 *d_bounds = nullptr;    // nranks+1
+    i64 *d_bounds = nullptr;     // nranks+1
     unsigned long long *d_count = nullptr;
     void *d_cub_tmp = nullptr;
     size_t cub_tmp_bytes = 0;
@@ -470,6 +562,7 @@ struct mv_engine {
     i64 *trace_target = nullptr;
     double *trace_mod = nullptr;
     int trace_cap = 0;
+    i64 *d_trace_tmp = nullptr;
 
     mv_stats stats{};
     std::vector<hipEvent_t> ev_pool;
@@ -529,14 +622,15 @@ void mv_engine_destroy(mv_engine *e) {
     hipSetDevice(e->device);
     for (auto ev : e->ev_pool) hipEventDestroy(ev);
     if (e->comm) ncclCommDestroy(e->comm);
-    // device allocations are freed with the primary context teardown at
-    // process exit; free the big ones explicitly
     for (void *p : {(void *)e->d_xadj, (void *)e->d_tails, (void *)e->d_ew,
-                    (void *)e->d_edges, (void *)e->d_curr, (void *)e->d_past,
-                    (void *)e->d_target, (void *)e->d_vdeg, (void *)e->d_cw,
-                    (void *)e->d_cinfo_size, (void *)e->d_cinfo_deg,
-                    (void *)e->d_cupd_size, (void *)e->d_cupd_deg,
-                    (void *)e->d_spill_k, (void *)e->d_spill_a})
+                    (void *)e->d_sigma, (void *)e->d_sigma_inv,
+                    (void *)e->d_sell_tidx, (void *)e->d_sell_w,
+                    (void *)e->d_deg, (void *)e->d_iota, (void *)e->d_perm,
+                    (void *)e->d_chunk_off, (void *)e->d_curr,
+                    (void *)e->d_past, (void *)e->d_target, (void *)e->d_vdeg,
+                    (void *)e->d_cw, (void *)e->d_cinfo, (void *)e->d_cupd,
+                    (void *)e->d_spill_k, (void *)e->d_spill_a,
+                    (void *)e->d_trace_tmp})
         if (p) hipFree(p);
     if (e->stream) hipStreamDestroy(e->stream);
     delete e;
@@ -571,31 +665,48 @@ int mv_engine_load_graph(mv_engine *e, const mv_graph *g) {
                         hipMemcpyHostToDevice));
     HIP_CHECK(hipMalloc(&e->d_ew, 8 * std::max<i64>(lne, 1)));
     HIP_CHECK(hipMemcpy(e->d_ew, w, 8 * lne, hipMemcpyHostToDevice));
-    HIP_CHECK(hipMalloc(&e->d_edges, sizeof(DevEdge) * std::max<i64>(lne, 1)));
+
+    // internal layout: the builder's spatial hint, or identity (the
+    // degree-sort fallback then runs in the per-run setup)
+    HIP_CHECK(hipMalloc(&e->d_sigma, 4 * std::max<i64>(lnv, 1)));
+    HIP_CHECK(hipMalloc(&e->d_sigma_inv, 4 * std::max<i64>(lnv, 1)));
+    const int32_t *hint = mv_graph_locality_hint(g);
+    e->has_hint = hint != nullptr;
+    if (hint) {
+        HIP_CHECK(hipMemcpy(e->d_sigma, hint, 4 * lnv, hipMemcpyHostToDevice));
+    } else {
+        k_iota32<<<grid_for(lnv), 256>>>(lnv, e->d_sigma);
+    }
+    k_invert_perm<<<grid_for(lnv), 256>>>(lnv, e->d_sigma, e->d_sigma_inv);
+
+    e->nchunks = (lnv + 63) / 64;
+    HIP_CHECK(hipMalloc(&e->d_deg, 4 * std::max<i64>(lnv, 1)));
+    HIP_CHECK(hipMalloc(&e->d_iota, 4 * std::max<i64>(lnv, 1)));
+    HIP_CHECK(hipMalloc(&e->d_perm, 4 * std::max<i64>(lnv, 1)));
+    HIP_CHECK(hipMalloc(&e->d_chunk_off, 8 * (e->nchunks + 1)));
 
     HIP_CHECK(hipMalloc(&e->d_curr, 8 * lnv));
     HIP_CHECK(hipMalloc(&e->d_past, 8 * lnv));
     HIP_CHECK(hipMalloc(&e->d_target, 8 * lnv));
     HIP_CHECK(hipMalloc(&e->d_vdeg, 8 * lnv));
     HIP_CHECK(hipMalloc(&e->d_cw, 8 * lnv));
-    HIP_CHECK(hipMalloc(&e->d_cinfo_size, 8 * lnv));
-    HIP_CHECK(hipMalloc(&e->d_cinfo_deg, 8 * lnv));
-    HIP_CHECK(hipMalloc(&e->d_cupd_size, 8 * lnv));
-    HIP_CHECK(hipMalloc(&e->d_cupd_deg, 8 * lnv));
+    HIP_CHECK(hipMalloc(&e->d_cinfo, sizeof(Cinfo) * lnv));
+    HIP_CHECK(hipMalloc(&e->d_cupd, sizeof(Cinfo) * lnv));
     HIP_CHECK(hipMalloc(&e->d_count, 8));
     HIP_CHECK(hipMalloc(&e->d_bounds, 8 * (e->nranks + 1)));
     HIP_CHECK(hipMalloc(&e->d_red, 16));
     const int nblocks = grid_for(lnv);
     HIP_CHECK(hipMalloc(&e->d_partials, 16 * nblocks));
 
-    // K4 geometry: 256-thread blocks, 16 LDS slots/lane (64 KiB/block ->
-    // 2 blocks/CU), spill region covers the max degree
+    // K4 geometry: 256-thread blocks, 8 LDS slots/lane (32 KiB/block ->
+    // 4-5 blocks/CU), spill region covers the max degree
     e->sweep_grid = grid_for(lnv, 256, 2048);
-    e->spill_max = (int)std::max<i64>(e->max_degree - 16 + 1, 1);
+    e->spill_max = (int)std::max<i64>(e->max_degree, 1); // covers any SLOTS
     const i64 nthreads = (i64)e->sweep_grid * 256;
     HIP_CHECK(hipMalloc(&e->d_spill_k, 8 * nthreads * e->spill_max));
     HIP_CHECK(hipMalloc(&e->d_spill_a, 8 * nthreads * e->spill_max));
 
+    HIP_CHECK(hipDeviceSynchronize());
     e->stats = mv_stats{};
     e->stats.edges_local = lne;
     return 0;
@@ -606,11 +717,13 @@ void mv_engine_set_trace(mv_engine *e, int64_t *target_trace, double *mod_trace,
     e->trace_target = target_trace;
     e->trace_mod = mod_trace;
     e->trace_cap = cap;
+    if (target_trace && !e->d_trace_tmp)
+        HIP_CHECK(hipMalloc(&e->d_trace_tmp, 8 * std::max<i64>(e->lnv, 1)));
 }
 
 void mv_engine_get_stats(const mv_engine *e, mv_stats *out) { *out = e->stats; }
 
-// helper: alltoallv over RCCL grouped send/recv; counts/displs in elements
+// helper: alltoallv over RCCL grouped send/recv; offsets in elements
 static void rccl_alltoallv(mv_engine *e, const void *send, const i64 *soff,
                            void *recv, const i64 *roff, size_t elem_bytes,
                            ncclDataType_t ty, size_t ty_bytes) {
@@ -718,11 +831,16 @@ double mv_engine_run(mv_engine *e, double lower, double thresh,
                     e->send_off[r] + ((r == me) ? 0 : matrix[(size_t)r * p + me]);
             e->ssz = e->send_off[p];
             if (e->d_svdata) HIP_CHECK(hipFree(e->d_svdata));
+            if (e->d_svdata_int) HIP_CHECK(hipFree(e->d_svdata_int));
             HIP_CHECK(hipMalloc(&e->d_svdata, 8 * std::max<i64>(e->ssz, 1)));
+            HIP_CHECK(hipMalloc(&e->d_svdata_int,
+                                4 * std::max<i64>(e->ssz, 1)));
             // role swap (dspl.hpp:1255-1257): my ghost list goes OUT, the
             // peers' lists land in svdata
             rccl_alltoallv(e, e->d_ghosts, e->recv_off.data(), e->d_svdata,
                            e->send_off.data(), 8, ncclInt64, 8);
+            k_to_internal<<<grid_for(std::max<i64>(e->ssz, 1)), 256, 0, st>>>(
+                e->ssz, e->d_svdata, e->base, e->d_sigma_inv, e->d_svdata_int);
             HIP_CHECK(hipStreamSynchronize(st));
 
             if (e->d_ghost_comm) HIP_CHECK(hipFree(e->d_ghost_comm));
@@ -734,15 +852,74 @@ double mv_engine_run(mv_engine *e, double lower, double thresh,
             e->nghost = 0;
             e->ssz = 0;
         }
-        // translate tails (K10b)
-        k_build_edges<<<grid_for(lne), 256, 0, st>>>(
-            lne, e->d_tails, e->d_ew, e->base, e->bound, lnv, e->d_ghosts,
-            e->nghost, e->d_edges);
+
+        // build the SELL image over the internal order (tails translated
+        // inline). perm = identity with a spatial hint, degree-sorted
+        // otherwise (load balance for skewed degree distributions).
+        k_degrees<<<grid_for(lnv), 256, 0, st>>>(lnv, e->d_sigma, e->d_xadj,
+                                                 e->d_deg);
+        if (e->has_hint) {
+            k_iota32<<<grid_for(lnv), 256, 0, st>>>(lnv, e->d_perm);
+        } else {
+            k_iota32<<<grid_for(lnv), 256, 0, st>>>(lnv, e->d_iota);
+            unsigned *d_degs = nullptr;
+            HIP_CHECK(hipMalloc(&d_degs, 4 * std::max<i64>(lnv, 1)));
+            size_t tb = 0;
+            hipcub::DeviceRadixSort::SortPairsDescending(
+                nullptr, tb, e->d_deg, d_degs, e->d_iota, e->d_perm, lnv, 0,
+                32, st);
+            void *d_tmp = nullptr;
+            HIP_CHECK(hipMalloc(&d_tmp, std::max<size_t>(tb, 1)));
+            hipcub::DeviceRadixSort::SortPairsDescending(
+                d_tmp, tb, e->d_deg, d_degs, e->d_iota, e->d_perm, lnv, 0, 32,
+                st);
+            HIP_CHECK(hipStreamSynchronize(st));
+            HIP_CHECK(hipFree(d_tmp));
+            HIP_CHECK(hipFree(d_degs));
+        }
+        {
+            i64 *d_sizes = nullptr;
+            HIP_CHECK(hipMalloc(&d_sizes, 8 * std::max<i64>(e->nchunks, 1)));
+            k_chunk_sizes<<<grid_for(e->nchunks), 256, 0, st>>>(
+                e->nchunks, lnv, e->d_perm, e->d_deg, d_sizes);
+            HIP_CHECK(hipMemsetAsync(e->d_chunk_off, 0, 8, st));
+            size_t tb2 = 0;
+            hipcub::DeviceScan::InclusiveSum(nullptr, tb2, d_sizes,
+                                             e->d_chunk_off + 1, e->nchunks,
+                                             st);
+            void *d_tmp2 = nullptr;
+            HIP_CHECK(hipMalloc(&d_tmp2, std::max<size_t>(tb2, 1)));
+            hipcub::DeviceScan::InclusiveSum(d_tmp2, tb2, d_sizes,
+                                             e->d_chunk_off + 1, e->nchunks,
+                                             st);
+            i64 total = 0;
+            HIP_CHECK(hipMemcpyAsync(&total, e->d_chunk_off + e->nchunks, 8,
+                                     hipMemcpyDeviceToHost, st));
+            HIP_CHECK(hipStreamSynchronize(st));
+            if (total > e->sell_elems) {
+                if (e->d_sell_tidx) HIP_CHECK(hipFree(e->d_sell_tidx));
+                if (e->d_sell_w) HIP_CHECK(hipFree(e->d_sell_w));
+                e->sell_elems = total;
+                HIP_CHECK(hipMalloc(&e->d_sell_tidx,
+                                    4 * std::max<i64>(total, 1)));
+                if (!e->unit_weights)
+                    HIP_CHECK(hipMalloc(&e->d_sell_w,
+                                        8 * std::max<i64>(total, 1)));
+            }
+            k_fill_sell<<<grid_for(lnv), 256, 0, st>>>(
+                lnv, e->d_perm, e->d_sigma, e->d_sigma_inv, e->d_xadj,
+                e->d_tails, e->d_ew, e->base, e->bound, e->d_ghosts, e->nghost,
+                e->d_chunk_off, e->d_sell_tidx,
+                e->unit_weights ? nullptr : e->d_sell_w);
+            HIP_CHECK(hipFree(d_tmp2));
+            HIP_CHECK(hipFree(d_sizes));
+        }
     }
 
     // ---- distInitLouvain (dspl.hpp:151-172) ----
     k1_vertex_degree<<<grid_for(lnv), 256, 0, st>>>(
-        lnv, e->d_xadj, e->d_ew, e->d_vdeg, e->d_cinfo_size, e->d_cinfo_deg);
+        lnv, e->d_sigma, e->d_xadj, e->d_ew, e->unit_weights, e->d_vdeg,
+        e->d_cinfo);
     const int nblocks = grid_for(lnv);
     {
         auto f = [vd = e->d_vdeg] __device__(i64 i, double &a, double &b) {
@@ -766,8 +943,8 @@ double mv_engine_run(mv_engine *e, double lower, double thresh,
         HIP_CHECK(hipStreamSynchronize(st));
     }
     const double constant = 1.0 / totalW; // dspl.hpp:129
-    k3_init_comm<<<grid_for(lnv), 256, 0, st>>>(lnv, e->base, e->d_curr,
-                                                e->d_past);
+    k3_init_comm<<<grid_for(lnv), 256, 0, st>>>(lnv, e->base, e->d_sigma,
+                                                e->d_curr, e->d_past);
     HIP_CHECK(hipStreamSynchronize(st));
     e->stats.setup_ms =
         std::chrono::duration<double, std::milli>(
@@ -788,7 +965,7 @@ double mv_engine_run(mv_engine *e, double lower, double thresh,
             const auto t_h0 = std::chrono::steady_clock::now();
             // ---- halo #1a: ghost communities (dspl.hpp:583-647) ----
             k8_gather_comms<<<grid_for(std::max<i64>(e->ssz, 1)), 256, 0, st>>>(
-                e->ssz, e->d_svdata, e->base, d_curr, e->d_scdata);
+                e->ssz, e->d_svdata_int, d_curr, e->d_scdata);
             rccl_alltoallv(e, e->d_scdata, e->send_off.data(), e->d_ghost_comm,
                            e->recv_off.data(), 8, ncclInt64, 8);
 
@@ -796,19 +973,15 @@ double mv_engine_run(mv_engine *e, double lower, double thresh,
             const i64 cand_max = e->nghost + lnv;
             if (cand_max > e->rc_cap) {
                 for (void *q : {(void *)e->d_cand, (void *)e->d_cand_sorted,
-                                (void *)e->d_rc_ids, (void *)e->d_rc_size,
-                                (void *)e->d_rc_degree, (void *)e->d_rcu_size,
-                                (void *)e->d_rcu_degree, (void *)e->d_rc_info})
+                                (void *)e->d_rc_ids, (void *)e->d_rc_info,
+                                (void *)e->d_rcu})
                     if (q) HIP_CHECK(hipFree(q));
                 e->rc_cap = cand_max;
                 HIP_CHECK(hipMalloc(&e->d_cand, 8 * cand_max));
                 HIP_CHECK(hipMalloc(&e->d_cand_sorted, 8 * cand_max));
                 HIP_CHECK(hipMalloc(&e->d_rc_ids, 8 * cand_max));
-                HIP_CHECK(hipMalloc(&e->d_rc_size, 8 * cand_max));
-                HIP_CHECK(hipMalloc(&e->d_rc_degree, 8 * cand_max));
-                HIP_CHECK(hipMalloc(&e->d_rcu_size, 8 * cand_max));
-                HIP_CHECK(hipMalloc(&e->d_rcu_degree, 8 * cand_max));
                 HIP_CHECK(hipMalloc(&e->d_rc_info, sizeof(Info16) * cand_max));
+                HIP_CHECK(hipMalloc(&e->d_rcu, sizeof(Info16) * cand_max));
                 size_t t1 = 0, t2 = 0;
                 hipcub::DeviceRadixSort::SortKeys(nullptr, t1, e->d_cand,
                                                   e->d_cand_sorted, cand_max, 0,
@@ -871,17 +1044,12 @@ double mv_engine_run(mv_engine *e, double lower, double thresh,
             rccl_alltoallv(e, e->d_rc_ids, rc_bounds.data(), e->d_req_ids,
                            req_off.data(), 8, ncclInt64, 8);
             k9_reply_info<<<grid_for(std::max<i64>(nreq, 1)), 256, 0, st>>>(
-                nreq, e->d_req_ids, e->base, e->d_cinfo_size, e->d_cinfo_deg,
-                e->d_req_info);
+                nreq, e->d_req_ids, e->base, e->d_cinfo, e->d_req_info);
             rccl_alltoallv(e, e->d_req_info, req_off.data(), e->d_rc_info,
-                           rc_bounds.data(), sizeof(Info16), ncclChar,
-                           1);
-            k_unpack_info<<<grid_for(std::max<i64>(nrc, 1)), 256, 0, st>>>(
-                nrc, e->d_rc_info, e->d_rc_size, e->d_rc_degree);
-            HIP_CHECK(hipMemsetAsync(e->d_rcu_size, 0, 8 * std::max<i64>(nrc, 1),
+                           rc_bounds.data(), sizeof(Info16), ncclChar, 1);
+            HIP_CHECK(hipMemsetAsync(e->d_rcu, 0,
+                                     sizeof(Info16) * std::max<i64>(nrc, 1),
                                      st));
-            HIP_CHECK(hipMemsetAsync(e->d_rcu_degree, 0,
-                                     8 * std::max<i64>(nrc, 1), st));
             HIP_CHECK(hipStreamSynchronize(st));
             e->stats.halo_ms +=
                 std::chrono::duration<double, std::milli>(
@@ -891,38 +1059,69 @@ double mv_engine_run(mv_engine *e, double lower, double thresh,
 
         // ---- K5 zero + K4 sweep (dspl.hpp:1371-1387) ----
         HIP_CHECK(hipMemsetAsync(e->d_cw, 0, 8 * lnv, st));
-        HIP_CHECK(hipMemsetAsync(e->d_cupd_size, 0, 8 * lnv, st));
-        HIP_CHECK(hipMemsetAsync(e->d_cupd_deg, 0, 8 * lnv, st));
+        HIP_CHECK(hipMemsetAsync(e->d_cupd, 0, sizeof(Cinfo) * lnv, st));
         hipEvent_t ev0 = e->ev_pair(), ev1 = e->ev_pair();
         HIP_CHECK(hipEventRecord(ev0, st));
-        constexpr int SLOTS = 16;
-        k4_sweep<SLOTS><<<e->sweep_grid, 256, SLOTS * 256 * 16, st>>>(
-            lnv, e->base, e->bound, e->d_xadj, e->d_edges, d_curr,
-            e->d_ghost_comm, e->d_vdeg, e->d_cinfo_size, e->d_cinfo_deg,
-            e->d_cupd_size, e->d_cupd_deg, e->d_rc_ids, nrc, e->d_rc_size,
-            e->d_rc_degree, e->d_rcu_size, e->d_rcu_degree, constant, d_target,
-            e->d_cw, e->d_spill_k, e->d_spill_a, e->spill_max);
+        // LDS slots per lane: early iterations see ~degree distinct
+        // candidate communities (every vertex its own community), later
+        // ones only a handful — a larger-slot / lower-occupancy variant for
+        // the first iterations avoids the spill path where it matters.
+        // Tunables (perf only, results identical): MV_SLOTS, MV_SLOTS_FIRST,
+        // MV_FIRST_ITERS.
+        static const int slots_rest = [] {
+            const char *s = getenv("MV_SLOTS");
+            return s ? atoi(s) : 8;
+        }();
+        static const int slots_first = [] {
+            const char *s = getenv("MV_SLOTS_FIRST");
+            return s ? atoi(s) : slots_rest;
+        }();
+        static const int first_iters = [] {
+            const char *s = getenv("MV_FIRST_ITERS");
+            return s ? atoi(s) : 2;
+        }();
+        const int slots = (numIters <= first_iters) ? slots_first : slots_rest;
+        auto launch_sweep = [&](auto slots_tag, auto unit_tag) {
+            constexpr int S = decltype(slots_tag)::value;
+            k4_sweep<S, decltype(unit_tag)::value>
+                <<<e->sweep_grid, 256, S * 256 * 16, st>>>(
+                    lnv, e->base, e->bound, e->d_perm, e->d_deg,
+                    e->d_chunk_off, e->d_sell_tidx, e->d_sell_w, d_curr,
+                    e->d_ghost_comm, e->d_vdeg, e->d_sigma, e->d_cinfo,
+                    e->d_cupd, e->d_rc_ids, nrc, e->d_rc_info, e->d_rcu,
+                    constant, d_target, e->d_cw, e->d_spill_k, e->d_spill_a,
+                    e->spill_max);
+        };
+        auto dispatch_slots = [&](auto unit_tag) {
+            switch (slots) {
+            case 4: launch_sweep(std::integral_constant<int, 4>{}, unit_tag); break;
+            case 12: launch_sweep(std::integral_constant<int, 12>{}, unit_tag); break;
+            case 16: launch_sweep(std::integral_constant<int, 16>{}, unit_tag); break;
+            case 24: launch_sweep(std::integral_constant<int, 24>{}, unit_tag); break;
+            default: launch_sweep(std::integral_constant<int, 8>{}, unit_tag); break;
+            }
+        };
+        if (e->unit_weights)
+            dispatch_slots(std::integral_constant<bool, true>{});
+        else
+            dispatch_slots(std::integral_constant<bool, false>{});
         HIP_CHECK(hipEventRecord(ev1, st));
         sweep_ev.push_back(ev0);
         sweep_ev.push_back(ev1);
         e->stats.sweep_launches++;
 
         // ---- K6 (dspl.hpp:458-471) ----
-        k6_apply_local<<<grid_for(lnv), 256, 0, st>>>(
-            lnv, e->d_cupd_size, e->d_cupd_deg, e->d_cinfo_size,
-            e->d_cinfo_deg);
+        k6_apply_local<<<grid_for(lnv), 256, 0, st>>>(lnv, e->d_cupd,
+                                                      e->d_cinfo);
 
         // ---- halo #2: route deltas to owners (dspl.hpp:978-1103) ----
         if (p > 1) {
             const auto t_h0 = std::chrono::steady_clock::now();
-            k_pack_deltas<<<grid_for(std::max<i64>(nrc, 1)), 256, 0, st>>>(
-                nrc, e->d_rcu_size, e->d_rcu_degree, e->d_rc_info);
-            rccl_alltoallv(e, e->d_rc_info, rc_bounds.data(), e->d_req_info,
+            rccl_alltoallv(e, e->d_rcu, rc_bounds.data(), e->d_req_info,
                            req_off.data(), sizeof(Info16), ncclChar, 1);
             k_apply_deltas<<<grid_for(std::max<i64>(req_off[p], 1)), 256, 0,
                              st>>>(req_off[p], e->d_req_ids, e->base,
-                                   e->d_req_info, e->d_cinfo_size,
-                                   e->d_cinfo_deg);
+                                   e->d_req_info, e->d_cinfo);
             HIP_CHECK(hipStreamSynchronize(st));
             e->stats.halo_ms +=
                 std::chrono::duration<double, std::milli>(
@@ -932,10 +1131,11 @@ double mv_engine_run(mv_engine *e, double lower, double thresh,
 
         // ---- K7: modularity (dspl.hpp:407-456) ----
         {
-            auto f = [cw = e->d_cw, cd = e->d_cinfo_deg] __device__(
+            auto f = [cw = e->d_cw, ci = e->d_cinfo] __device__(
                          i64 i, double &a, double &b) {
                 a = cw[i];
-                b = cd[i] * cd[i];
+                const double d = ci[i].degree;
+                b = d * d;
             };
             k_partial_sum2<<<nblocks, 256, 0, st>>>(lnv, f, e->d_partials);
         }
@@ -962,9 +1162,16 @@ double mv_engine_run(mv_engine *e, double lower, double thresh,
         // ---- trace ----
         if (e->trace_mod && numIters <= e->trace_cap)
             e->trace_mod[numIters - 1] = currMod;
-        if (e->trace_target && numIters <= e->trace_cap)
-            HIP_CHECK(hipMemcpy(e->trace_target + (i64)(numIters - 1) * lnv,
-                                d_target, 8 * lnv, hipMemcpyDeviceToHost));
+        if (e->trace_target && numIters <= e->trace_cap) {
+            k_depermute<<<grid_for(lnv), 256, 0, st>>>(lnv, e->d_sigma_inv,
+                                                       d_target,
+                                                       e->d_trace_tmp);
+            HIP_CHECK(hipMemcpyAsync(e->trace_target +
+                                         (i64)(numIters - 1) * lnv,
+                                     e->d_trace_tmp, 8 * lnv,
+                                     hipMemcpyDeviceToHost, st));
+            HIP_CHECK(hipStreamSynchronize(st));
+        }
 
         if (currMod - prevMod < thresh) break; // dspl.hpp:1401
         prevMod = currMod;
@@ -978,10 +1185,13 @@ double mv_engine_run(mv_engine *e, double lower, double thresh,
         if (numIters >= 10000) break; // safety net, never hit in practice
     }
 
+    const bool dbg = getenv("MV_SWEEP_DEBUG") != nullptr;
     for (size_t k = 0; k + 1 < sweep_ev.size() + 1; k += 2) {
         float ms = 0;
         HIP_CHECK(hipEventElapsedTime(&ms, sweep_ev[k], sweep_ev[k + 1]));
         e->stats.sweep_ms += ms;
+        if (dbg)
+            std::fprintf(stderr, "[sweep] iter %zu: %.3f ms\n", k / 2 + 1, ms);
     }
     e->stats.iters = numIters;
     e->stats.total_ms = std::chrono::duration<double, std::milli>(

@@ -80,6 +80,14 @@ struct mv_graph {
     std::vector<int64_t> xadj;  // lnv+1
     std::vector<int64_t> tails; // lne, global ids
     std::vector<double> weights;
+    // Optional INTERNAL-layout hint: locality_perm[k] = the original local
+    // vertex id placed at internal position k, ordered so that spatially
+    // close vertices (whose ids the RGG assigns randomly) sit close in
+    // memory. Pure layout metadata — labels, results and the wire protocol
+    // are untouched; the engine uses it so neighbor-community gathers hit
+    // L1/L2 instead of round-tripping to the Infinity Cache. Empty when the
+    // builder has no spatial knowledge (binary reader, from_csr).
+    std::vector<int32_t> locality_perm;
 };
 
 extern "C" {
@@ -257,6 +265,25 @@ mv_graph *mv_graph_rgg(int64_t nv, int rank, int nranks,
         }
     }
 
+    // spatial layout hint: own vertices in row-major cell order (stable, so
+    // ties keep id order) — see mv_graph::locality_perm
+    {
+        std::vector<std::pair<int64_t, int32_t>> order(n_);
+        for (int64_t i = 0; i < n_; i++) {
+            const double xi = X[own_off + i], yi = Y[own_off + i];
+            const int cx = std::min((int)(xi * inv_cs), ncx - 1);
+            const int cy = std::min((int)(yi * inv_cs), ncx - 1);
+            order[i] = {(int64_t)cy * ncx + cx, (int32_t)i};
+        }
+        std::stable_sort(order.begin(), order.end(),
+                         [](const auto &a, const auto &b) {
+                             return a.first < b.first;
+                         });
+        g->locality_perm.resize(n_);
+        for (int64_t k = 0; k < n_; k++)
+            g->locality_perm[k] = order[k].second;
+    }
+
     int64_t lne = 0;
     for (int64_t i = 0; i < n_; i++) {
         g->xadj[i + 1] = g->xadj[i] + (int64_t)rows[i].size();
@@ -405,5 +432,8 @@ const int64_t *mv_graph_parts(const mv_graph *g) { return g->parts.data(); }
 const int64_t *mv_graph_xadj(const mv_graph *g) { return g->xadj.data(); }
 const int64_t *mv_graph_tails(const mv_graph *g) { return g->tails.data(); }
 const double *mv_graph_weights(const mv_graph *g) { return g->weights.data(); }
+const int32_t *mv_graph_locality_hint(const mv_graph *g) {
+    return g->locality_perm.empty() ? nullptr : g->locality_perm.data();
+}
 
 } // extern "C"
