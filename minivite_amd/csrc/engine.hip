@@ -316,13 +316,16 @@ __global__ void k_apply_deltas(i64 n, const i64 *__restrict__ ids, i64 base,
     }
 }
 
-// ---- K6: apply localCupdate (dspl.hpp:458-471) ----
-__global__ void k6_apply_local(i64 lnv, const Cinfo *__restrict__ cupd,
+// ---- K6: apply localCupdate (dspl.hpp:458-471), fused with the next
+// iteration's K5 zeroing of localCupdate (dspl.hpp:483-484) ----
+__global__ void k6_apply_local(i64 lnv, Cinfo *__restrict__ cupd,
                                Cinfo *__restrict__ cinfo) {
     for (i64 i = blockIdx.x * (i64)blockDim.x + threadIdx.x; i < lnv;
          i += (i64)gridDim.x * blockDim.x) {
-        cinfo[i].size += cupd[i].size;
-        cinfo[i].degree += cupd[i].degree;
+        const Cinfo u = cupd[i];
+        cinfo[i].size += u.size;
+        cinfo[i].degree += u.degree;
+        cupd[i] = {0, 0.0};
     }
 }
 
@@ -383,76 +386,113 @@ __global__ __launch_bounds__(256) void k4_sweep(
         }
         i64 target;
         if (deg == 0) {
-            target = cc; // dspl.hpp:323-324
+            target = cc;          // dspl.hpp:323-324
+            clusterWeight[i] = 0; // K5 semantics (dspl.hpp:481-482)
         } else {
             double c0 = 0.0, selfLoop = 0.0;
             int ns = 0, nspill = 0;
-            for (int k = 0; k < deg; k++) {
-                const i64 tidx = sell_tidx[ebase + (i64)k * 64];
-                const double w = UNIT ? 1.0 : sell_w[ebase + (i64)k * 64];
-                if (tidx == i) selfLoop += w; // dspl.hpp:247-248
-                const i64 tcomm = (tidx < lnv) ? currComm[tidx]
-                                               : ghost_comm[tidx - lnv];
-                if (tcomm == cc) { c0 += w; continue; }
-                bool found = false;
-                for (int t = 0; t < ns; t++) {
-                    if (skey[t * blockDim.x + tid] == tcomm) {
-                        sacc[t * blockDim.x + tid] += w;
-                        found = true;
-                        break;
+            // Edges in chunks of 8: issue the 8 independent tail loads and
+            // the 8 dependent community gathers together (the per-edge
+            // serial load->gather->probe chain was the measured bound: 80%
+            // SQ_WAIT_ANY with LDS and VALU idle), then probe sequentially
+            // in edge order (bit-exact -w accumulation, dspl.hpp:240-271).
+            constexpr int CH = 8;
+            for (int k0 = 0; k0 < deg; k0 += CH) {
+                const int m = min(CH, deg - k0);
+                i64 tb[CH];
+                i64 cb[CH];
+                double wb[CH];
+#pragma unroll
+                for (int j = 0; j < CH; j++) {
+                    const i64 slot =
+                        (j < m) ? ebase + (i64)(k0 + j) * 64 : ebase;
+                    tb[j] = sell_tidx[slot];
+                    if (!UNIT) wb[j] = sell_w[slot];
+                }
+#pragma unroll
+                for (int j = 0; j < CH; j++)
+                    cb[j] = (tb[j] < lnv) ? currComm[tb[j]]
+                                          : ghost_comm[tb[j] - lnv];
+                for (int j = 0; j < m; j++) {
+                    const i64 tidx = tb[j];
+                    const double w = UNIT ? 1.0 : wb[j];
+                    if (tidx == i) selfLoop += w; // dspl.hpp:247-248
+                    const i64 tcomm = cb[j];
+                    if (tcomm == cc) { c0 += w; continue; }
+                    bool found = false;
+                    for (int t = 0; t < ns; t++) {
+                        if (skey[t * blockDim.x + tid] == tcomm) {
+                            sacc[t * blockDim.x + tid] += w;
+                            found = true;
+                            break;
+                        }
                     }
-                }
-                if (found) continue;
-                if (ns < SLOTS) {
-                    skey[ns * blockDim.x + tid] = tcomm;
-                    sacc[ns * blockDim.x + tid] = w;
-                    ns++;
-                    continue;
-                }
-                for (int t = 0; t < nspill; t++) {
-                    if (myspill_k[t] == tcomm) {
-                        myspill_a[t] += w;
-                        found = true;
-                        break;
+                    if (found) continue;
+                    if (ns < SLOTS) {
+                        skey[ns * blockDim.x + tid] = tcomm;
+                        sacc[ns * blockDim.x + tid] = w;
+                        ns++;
+                        continue;
                     }
-                }
-                if (!found) { // spill_max covers the max degree: no overflow
-                    myspill_k[nspill] = tcomm;
-                    myspill_a[nspill] = w;
-                    nspill++;
+                    for (int t = 0; t < nspill; t++) {
+                        if (myspill_k[t] == tcomm) {
+                            myspill_a[t] += w;
+                            found = true;
+                            break;
+                        }
+                    }
+                    if (!found) { // spill_max covers the max degree
+                        myspill_k[nspill] = tcomm;
+                        myspill_a[nspill] = w;
+                        nspill++;
+                    }
                 }
             }
-            clusterWeight[i] += c0; // dspl.hpp:318
+            clusterWeight[i] = c0; // dspl.hpp:318 onto the K5-zeroed value
 
-            // distGetMaxIndex (dspl.hpp:174-228)
+            // distGetMaxIndex (dspl.hpp:174-228); candidate info gathers
+            // batched 8-wide like the edge loop (independent 16-B loads)
             const double vdeg = vDegree[i];
             const double eix = c0 - selfLoop;
             const double ax = ccDeg - vdeg;
             double maxGain = 0.0;
             i64 maxIndex = cc, maxSize = ccSize;
-            for (int t = 0; t < ns + nspill; t++) {
-                const i64 y = (t < ns) ? skey[t * blockDim.x + tid]
-                                       : myspill_k[t - ns];
-                const double eiy = (t < ns) ? sacc[t * blockDim.x + tid]
-                                            : myspill_a[t - ns];
-                double ay;
-                i64 ysz;
-                if (y >= base && y < bound) {
-                    const Cinfo c = cinfo[y - base];
-                    ay = c.degree;
-                    ysz = c.size;
-                } else {
-                    const i64 q = dev_bsearch(rc_ids, nrc, y);
-                    ay = rc_info[q].degree;
-                    ysz = rc_info[q].size;
+            const int tot = ns + nspill;
+            for (int t0 = 0; t0 < tot; t0 += CH) {
+                const int m2 = min(CH, tot - t0);
+                i64 yb[CH];
+                double eb[CH], ab[CH];
+                i64 zb[CH];
+#pragma unroll
+                for (int j = 0; j < CH; j++) {
+                    const int t = (j < m2) ? t0 + j : t0;
+                    yb[j] = (t < ns) ? skey[t * blockDim.x + tid]
+                                     : myspill_k[t - ns];
+                    eb[j] = (t < ns) ? sacc[t * blockDim.x + tid]
+                                     : myspill_a[t - ns];
                 }
-                const double curGain =
-                    2.0 * (eiy - eix) - 2.0 * vdeg * (ay - ax) * constant; // :212
-                if (curGain > maxGain ||
-                    (curGain == maxGain && curGain != 0.0 && y < maxIndex)) {
-                    maxGain = curGain;
-                    maxIndex = y;
-                    maxSize = ysz;
+#pragma unroll
+                for (int j = 0; j < CH; j++) {
+                    if (yb[j] >= base && yb[j] < bound) {
+                        const Cinfo c = cinfo[yb[j] - base];
+                        ab[j] = c.degree;
+                        zb[j] = c.size;
+                    } else {
+                        const i64 q = dev_bsearch(rc_ids, nrc, yb[j]);
+                        ab[j] = rc_info[q].degree;
+                        zb[j] = rc_info[q].size;
+                    }
+                }
+                for (int j = 0; j < m2; j++) {
+                    const double curGain = 2.0 * (eb[j] - eix) -
+                                           2.0 * vdeg * (ab[j] - ax) * constant; // :212
+                    if (curGain > maxGain ||
+                        (curGain == maxGain && curGain != 0.0 &&
+                         yb[j] < maxIndex)) {
+                        maxGain = curGain;
+                        maxIndex = yb[j];
+                        maxSize = zb[j];
+                    }
                 }
             }
             if (maxSize == 1 && ccSize == 1 && maxIndex > cc) // :224-225
@@ -485,6 +525,106 @@ __global__ __launch_bounds__(256) void k4_sweep(
     }
 }
 
+// ---- K4 first-iteration specialization ----
+// At iteration 1 currComm is the identity (dspl.hpp:145-147), so per vertex:
+// counter[0] collects exactly the self-loop weight (eix = counter[0] -
+// selfLoop == 0.0, dspl.hpp:183), ax = ccDegree - vDegree == 0.0 (:185),
+// every community has size 1, and each distinct neighbor TAIL is its own
+// candidate. With per-row tails sorted ascending (the reference CSR's
+// order, graph.hpp:1145-1153 — verified at load), parallel edges are
+// adjacent and candidates arrive in ascending label order, so the
+// tie-break (max gain, then smallest label, dspl.hpp:214-215) reduces to a
+// strict-greater streaming argmax: no clmap at all, and the candidate's
+// degree ay is vDegree[tidx] — a spatially local gather instead of a
+// random cinfo line. The singleton guard (dspl.hpp:224-225) is the final
+// maxIndex > cc check. Gains carry the identical bits: 2.0*(eiy-0.0) and
+// (ay-0.0) round exactly like the reference's expressions.
+template <bool UNIT>
+__global__ __launch_bounds__(256) void k4_sweep_iter1(
+    i64 lnv, i64 base, i64 bound, const unsigned *__restrict__ perm,
+    const unsigned *__restrict__ deg_int, const i64 *__restrict__ chunk_off,
+    const int *__restrict__ sell_tidx, const double *__restrict__ sell_w,
+    const i64 *__restrict__ currComm, const i64 *__restrict__ ghost_comm,
+    const double *__restrict__ vDegree, Cinfo *__restrict__ cupd,
+    const i64 *__restrict__ rc_ids, i64 nrc,
+    const Info16 *__restrict__ rc_info, Info16 *__restrict__ rcu,
+    double constant, i64 *__restrict__ targetComm,
+    double *__restrict__ clusterWeight) {
+    const i64 gthread = blockIdx.x * (i64)blockDim.x + threadIdx.x;
+    const i64 stride = (i64)gridDim.x * blockDim.x;
+    for (i64 s = gthread; s < lnv; s += stride) {
+        const i64 i = perm[s];
+        const int deg = (int)deg_int[i];
+        const i64 ebase = chunk_off[s >> 6] + (s & 63);
+        const i64 cc = currComm[i]; // own label
+        const double vdeg = vDegree[i];
+        double c0 = 0.0;
+        double maxGain = 0.0;
+        i64 maxIndex = cc;
+        i64 prev = INT64_MIN;
+        i64 pend_label = 0;
+        double eiy = 0.0, pend_ay = 0.0;
+        bool pend = false;
+        constexpr int CH = 8;
+        for (int k0 = 0; k0 < deg; k0 += CH) {
+            const int m = min(CH, deg - k0);
+            i64 tb[CH], cb[CH];
+            double wb[CH], vb[CH];
+#pragma unroll
+            for (int j = 0; j < CH; j++) {
+                const i64 slot = (j < m) ? ebase + (i64)(k0 + j) * 64 : ebase;
+                tb[j] = sell_tidx[slot];
+                if (!UNIT) wb[j] = sell_w[slot];
+            }
+#pragma unroll
+            for (int j = 0; j < CH; j++) {
+                cb[j] = (tb[j] < lnv) ? currComm[tb[j]]
+                                      : ghost_comm[tb[j] - lnv];
+                vb[j] = (tb[j] < lnv) ? vDegree[tb[j]] : 0.0;
+            }
+            for (int j = 0; j < m; j++) {
+                const i64 tidx = tb[j];
+                const double w = UNIT ? 1.0 : wb[j];
+                if (tidx == i) { c0 += w; continue; } // self: counter[0]
+                if (tidx == prev) { eiy += w; continue; } // parallel edge
+                if (pend) {
+                    const double g =
+                        2.0 * eiy - 2.0 * vdeg * pend_ay * constant;
+                    if (g > maxGain) { maxGain = g; maxIndex = pend_label; }
+                }
+                prev = tidx;
+                pend_label = cb[j];
+                eiy = w;
+                pend_ay = (tidx < lnv)
+                              ? vb[j]
+                              : rc_info[dev_bsearch(rc_ids, nrc, cb[j])].degree;
+                pend = true;
+            }
+        }
+        if (pend) {
+            const double g = 2.0 * eiy - 2.0 * vdeg * pend_ay * constant;
+            if (g > maxGain) { maxGain = g; maxIndex = pend_label; }
+        }
+        if (maxIndex > cc) maxIndex = cc; // singleton guard: all sizes are 1
+        clusterWeight[i] = c0;            // dspl.hpp:318 (eix == 0)
+        if (maxIndex != cc) {             // cc is local at iteration 1
+            Cinfo *u = &cupd[cc - base];
+            atomicAdd(&u->degree, -vdeg);
+            atomic_add_i64(&u->size, -1);
+            if (maxIndex >= base && maxIndex < bound) {
+                Cinfo *t = &cupd[maxIndex - base];
+                atomicAdd(&t->degree, vdeg);
+                atomic_add_i64(&t->size, 1);
+            } else {
+                const i64 q = dev_bsearch(rc_ids, nrc, maxIndex);
+                atomicAdd(&rcu[q].degree, vdeg);
+                atomic_add_i64(&rcu[q].size, 1);
+            }
+        }
+        targetComm[i] = maxIndex;
+    }
+}
+
 int grid_for(i64 n, int block = 256, int cap = 2048) {
     i64 g = (n + block - 1) / block;
     return (int)std::min<i64>(std::max<i64>(g, 1), cap);
@@ -506,6 +646,7 @@ struct mv_engine {
     i64 *d_tails = nullptr;   // raw global tails (kept for setup)
     double *d_ew = nullptr;   // edge weights (CSR order)
     int unit_weights = 1;
+    int rows_sorted = 1; // per-row tails ascending (reference CSR order)
     i64 max_degree = 0;
 
     // internal layout
@@ -654,6 +795,11 @@ int mv_engine_load_graph(mv_engine *e, const mv_graph *g) {
     e->unit_weights = 1;
     for (i64 k = 0; k < lne; k++)
         if (w[k] != 1.0) { e->unit_weights = 0; break; }
+    const i64 *tails_h = mv_graph_tails(g);
+    e->rows_sorted = 1;
+    for (i64 i = 0; i < lnv && e->rows_sorted; i++)
+        for (i64 k = xadj[i] + 1; k < xadj[i + 1]; k++)
+            if (tails_h[k - 1] > tails_h[k]) { e->rows_sorted = 0; break; }
 
     HIP_CHECK(hipMalloc(&e->d_parts, 8 * (e->nranks + 1)));
     HIP_CHECK(hipMemcpy(e->d_parts, e->parts_h.data(), 8 * (e->nranks + 1),
@@ -1057,9 +1203,11 @@ double mv_engine_run(mv_engine *e, double lower, double thresh,
                     .count();
         }
 
-        // ---- K5 zero + K4 sweep (dspl.hpp:1371-1387) ----
-        HIP_CHECK(hipMemsetAsync(e->d_cw, 0, 8 * lnv, st));
-        HIP_CHECK(hipMemsetAsync(e->d_cupd, 0, sizeof(Cinfo) * lnv, st));
+        // ---- K4 sweep (dspl.hpp:1371-1387; K5's zeroing is fused: cupd is
+        // zeroed by K6 after each apply + once before the loop, and the
+        // sweep overwrites clusterWeight instead of accumulating) ----
+        if (numIters == 1)
+            HIP_CHECK(hipMemsetAsync(e->d_cupd, 0, sizeof(Cinfo) * lnv, st));
         hipEvent_t ev0 = e->ev_pair(), ev1 = e->ev_pair();
         HIP_CHECK(hipEventRecord(ev0, st));
         // LDS slots per lane: early iterations see ~degree distinct
@@ -1101,10 +1249,24 @@ double mv_engine_run(mv_engine *e, double lower, double thresh,
             default: launch_sweep(std::integral_constant<int, 8>{}, unit_tag); break;
             }
         };
-        if (e->unit_weights)
+        auto launch_iter1 = [&](auto unit_tag) {
+            k4_sweep_iter1<decltype(unit_tag)::value>
+                <<<e->sweep_grid, 256, 0, st>>>(
+                    lnv, e->base, e->bound, e->d_perm, e->d_deg,
+                    e->d_chunk_off, e->d_sell_tidx, e->d_sell_w, d_curr,
+                    e->d_ghost_comm, e->d_vdeg, e->d_cupd, e->d_rc_ids, nrc,
+                    e->d_rc_info, e->d_rcu, constant, d_target, e->d_cw);
+        };
+        if (numIters == 1 && e->rows_sorted) {
+            if (e->unit_weights)
+                launch_iter1(std::integral_constant<bool, true>{});
+            else
+                launch_iter1(std::integral_constant<bool, false>{});
+        } else if (e->unit_weights) {
             dispatch_slots(std::integral_constant<bool, true>{});
-        else
+        } else {
             dispatch_slots(std::integral_constant<bool, false>{});
+        }
         HIP_CHECK(hipEventRecord(ev1, st));
         sweep_ev.push_back(ev0);
         sweep_ev.push_back(ev1);
