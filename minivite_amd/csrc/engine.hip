@@ -329,6 +329,41 @@ __global__ void k6_apply_local(i64 lnv, Cinfo *__restrict__ cupd,
     }
 }
 
+// K6+K7 fused (single-rank path): apply localCupdate, zero it, and emit
+// this block's modularity partials in one pass (no remote deltas exist
+// between K6 and K7 when nranks == 1)
+__global__ void k67_apply_and_partials(i64 lnv, Cinfo *__restrict__ cupd,
+                                       Cinfo *__restrict__ cinfo,
+                                       const double *__restrict__ cw,
+                                       double *__restrict__ out2) {
+    double a = 0.0, b = 0.0;
+    for (i64 i = blockIdx.x * (i64)blockDim.x + threadIdx.x; i < lnv;
+         i += (i64)gridDim.x * blockDim.x) {
+        const Cinfo u = cupd[i];
+        const i64 nsz = cinfo[i].size + u.size;
+        const double nde = cinfo[i].degree + u.degree;
+        cinfo[i] = {nsz, nde};
+        cupd[i] = {0, 0.0};
+        a += cw[i];
+        b += nde * nde;
+    }
+    __shared__ double sa[256], sb[256];
+    sa[threadIdx.x] = a;
+    sb[threadIdx.x] = b;
+    __syncthreads();
+    for (int s = blockDim.x / 2; s > 0; s >>= 1) {
+        if (threadIdx.x < s) {
+            sa[threadIdx.x] += sa[threadIdx.x + s];
+            sb[threadIdx.x] += sb[threadIdx.x + s];
+        }
+        __syncthreads();
+    }
+    if (threadIdx.x == 0) {
+        out2[2 * blockIdx.x] = sa[0];
+        out2[2 * blockIdx.x + 1] = sb[0];
+    }
+}
+
 // de-permute an internal-ordered array into label order (trace only)
 __global__ void k_depermute(i64 lnv, const unsigned *__restrict__ sigma_inv,
                             const i64 *__restrict__ in,
@@ -1272,9 +1307,14 @@ double mv_engine_run(mv_engine *e, double lower, double thresh,
         sweep_ev.push_back(ev1);
         e->stats.sweep_launches++;
 
-        // ---- K6 (dspl.hpp:458-471) ----
-        k6_apply_local<<<grid_for(lnv), 256, 0, st>>>(lnv, e->d_cupd,
-                                                      e->d_cinfo);
+        // ---- K6 (dspl.hpp:458-471); fused with K7 at p==1 ----
+        if (p == 1) {
+            k67_apply_and_partials<<<nblocks, 256, 0, st>>>(
+                lnv, e->d_cupd, e->d_cinfo, e->d_cw, e->d_partials);
+        } else {
+            k6_apply_local<<<grid_for(lnv), 256, 0, st>>>(lnv, e->d_cupd,
+                                                          e->d_cinfo);
+        }
 
         // ---- halo #2: route deltas to owners (dspl.hpp:978-1103) ----
         if (p > 1) {
@@ -1291,8 +1331,9 @@ double mv_engine_run(mv_engine *e, double lower, double thresh,
                     .count();
         }
 
-        // ---- K7: modularity (dspl.hpp:407-456) ----
-        {
+        // ---- K7: modularity (dspl.hpp:407-456; p==1 already emitted the
+        // partials in the fused kernel above) ----
+        if (p > 1) {
             auto f = [cw = e->d_cw, ci = e->d_cinfo] __device__(
                          i64 i, double &a, double &b) {
                 a = cw[i];

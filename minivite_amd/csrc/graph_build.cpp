@@ -170,6 +170,13 @@ mv_graph *mv_graph_rgg(int64_t nv, int rank, int nranks,
         for (int64_t t = 0; t < npts; t++)
             cpts[cur[cell_of(X[t], Y[t])]++] = t;
     }
+    // coordinates packed in cell order: the pair loop below then streams
+    // contiguous memory instead of gathering across the whole strip
+    std::vector<double> Xc(npts), Yc(npts);
+    for (int64_t q = 0; q < npts; q++) {
+        Xc[q] = X[cpts[q]];
+        Yc[q] = Y[cpts[q]];
+    }
 
     // Row of own vertex i = every neighbor within rn in any strip:
     //  - own strip: all j != i (reference emits (i,g_j) and (j,g_i) per
@@ -184,16 +191,13 @@ mv_graph *mv_graph_rgg(int64_t nv, int rank, int nranks,
     const int64_t own_off = (int64_t)(rank - s_lo) * n_;
     const int64_t base = g->parts[rank];
 
+    // two passes (count, then fill + per-row insertion sort) — no
+    // intermediate per-row vectors, so n=2^26 fits comfortably
     g->xadj.assign(n_ + 1, 0);
-    struct TW { int64_t t; double w; };
-    std::vector<std::vector<TW>> rows(n_);
-
-#pragma omp parallel for schedule(dynamic, 1024)
-    for (int64_t i = 0; i < n_; i++) {
+    auto for_each_neighbor = [&](int64_t i, auto &&emit) {
         const double xi = X[own_off + i], yi = Y[own_off + i];
         const int cxi = std::min((int)(xi * inv_cs), ncx - 1);
         const int cyi = std::min((int)(yi * inv_cs), ncx - 1);
-        auto &row = rows[i];
         for (int dy = -1; dy <= 1; dy++) {
             const int cy = cyi + dy;
             if (cy < row0 || cy > row1) continue;
@@ -202,23 +206,47 @@ mv_graph *mv_graph_rgg(int64_t nv, int rank, int nranks,
                 if (cx < 0 || cx >= ncx) continue;
                 const int64_t c = (int64_t)(cy - row0) * ncx + cx;
                 for (int64_t q = ccount[c]; q < ccount[c + 1]; q++) {
+                    const double ddx = xi - Xc[q];
+                    const double ddy = yi - Yc[q];
+                    const double ed = std::sqrt(ddx * ddx + ddy * ddy);
+                    if (ed > rn) continue;
                     const int64_t t = cpts[q];
                     if (t == own_off + i) continue;
                     const int strip = (int)(t / n_) + s_lo;
                     const int64_t j = t % n_;
                     if (strip != rank && j == i) continue; // the j>i quirk
-                    const double ddx = xi - X[t];
-                    const double ddy = yi - Y[t];
-                    const double ed = std::sqrt(ddx * ddx + ddy * ddy);
-                    if (ed <= rn) {
-                        row.push_back({(int64_t)strip * n_ + j,
-                                       unit_edge_weight ? 1.0 : ed});
-                    }
+                    emit((int64_t)strip * n_ + j, ed);
                 }
             }
         }
-        std::sort(row.begin(), row.end(),
-                  [](const TW &a, const TW &b) { return a.t < b.t; });
+    };
+#pragma omp parallel for schedule(dynamic, 4096)
+    for (int64_t i = 0; i < n_; i++) {
+        int64_t cnt = 0;
+        for_each_neighbor(i, [&](int64_t, double) { cnt++; });
+        g->xadj[i + 1] = cnt;
+    }
+    for (int64_t i = 0; i < n_; i++) g->xadj[i + 1] += g->xadj[i];
+    const int64_t lne_pre = g->xadj[n_];
+    g->tails.resize(lne_pre);
+    g->weights.resize(lne_pre);
+#pragma omp parallel for schedule(dynamic, 4096)
+    for (int64_t i = 0; i < n_; i++) {
+        int64_t o = g->xadj[i];
+        const int64_t o0 = o;
+        for_each_neighbor(i, [&](int64_t t, double ed) {
+            // insertion in ascending tail order (rows are short)
+            int64_t k = o;
+            const double wv = unit_edge_weight ? 1.0 : ed;
+            while (k > o0 && g->tails[k - 1] > t) {
+                g->tails[k] = g->tails[k - 1];
+                g->weights[k] = g->weights[k - 1];
+                k--;
+            }
+            g->tails[k] = t;
+            g->weights[k] = wv;
+            o++;
+        });
     }
 
     // Optional extra random edges (mirrors -p, graph.hpp:939-1122, with an
@@ -237,6 +265,7 @@ mv_graph *mv_graph_rgg(int64_t nv, int rank, int nranks,
         const int64_t nrande =
             (int64_t)(random_edge_percent * (double)tot_und) / 100;
         int64_t per = nrande / nranks;
+        struct TW { int64_t t; double w; };
         std::vector<std::vector<TW>> extra(n_);
         for (int src_rank = 0; src_rank < nranks; src_rank++) {
             std::mt19937_64 re(random_edge_seed * 0x9E3779B97F4A7C15ull +
@@ -256,12 +285,38 @@ mv_graph *mv_graph_rgg(int64_t nv, int rank, int nranks,
                 if (target == rank) extra[j].push_back({g_i, 1.0});
             }
         }
-        for (int64_t i = 0; i < n_; i++) {
-            if (extra[i].empty()) continue;
-            auto &row = rows[i];
-            row.insert(row.end(), extra[i].begin(), extra[i].end());
-            std::sort(row.begin(), row.end(),
-                      [](const TW &a, const TW &b) { return a.t < b.t; });
+        // merge the extra edges into the CSR (rebuild with shifted rows)
+        int64_t n_extra = 0;
+        for (auto &v : extra) n_extra += (int64_t)v.size();
+        if (n_extra) {
+            std::vector<int64_t> nxadj(n_ + 1, 0);
+            for (int64_t i = 0; i < n_; i++)
+                nxadj[i + 1] = nxadj[i] + (g->xadj[i + 1] - g->xadj[i]) +
+                               (int64_t)extra[i].size();
+            std::vector<int64_t> ntails(nxadj[n_]);
+            std::vector<double> nweights(nxadj[n_]);
+#pragma omp parallel for schedule(static)
+            for (int64_t i = 0; i < n_; i++) {
+                int64_t o = nxadj[i];
+                for (int64_t e = g->xadj[i]; e < g->xadj[i + 1]; e++, o++) {
+                    ntails[o] = g->tails[e];
+                    nweights[o] = g->weights[e];
+                }
+                for (auto &x : extra[i]) {
+                    int64_t k = o;
+                    while (k > nxadj[i] && ntails[k - 1] > x.t) {
+                        ntails[k] = ntails[k - 1];
+                        nweights[k] = nweights[k - 1];
+                        k--;
+                    }
+                    ntails[k] = x.t;
+                    nweights[k] = x.w;
+                    o++;
+                }
+            }
+            g->xadj = std::move(nxadj);
+            g->tails = std::move(ntails);
+            g->weights = std::move(nweights);
         }
     }
 
@@ -284,22 +339,6 @@ mv_graph *mv_graph_rgg(int64_t nv, int rank, int nranks,
             g->locality_perm[k] = order[k].second;
     }
 
-    int64_t lne = 0;
-    for (int64_t i = 0; i < n_; i++) {
-        g->xadj[i + 1] = g->xadj[i] + (int64_t)rows[i].size();
-        lne += (int64_t)rows[i].size();
-    }
-    g->tails.resize(lne);
-    g->weights.resize(lne);
-#pragma omp parallel for schedule(static)
-    for (int64_t i = 0; i < n_; i++) {
-        int64_t o = g->xadj[i];
-        for (auto &e : rows[i]) {
-            g->tails[o] = e.t;
-            g->weights[o] = e.w;
-            o++;
-        }
-    }
     return g;
 }
 
