@@ -132,3 +132,30 @@ def test_multi_gpu_parity(world):
             tr = np.frombuffer(results[r][2], dtype=np.int64).reshape(iters, -1)
             full[parts[r]:parts[r + 1]] = tr[k]
         assert sha(full) == pin["iter_target_sha"][k], f"iteration {k+1}"
+
+
+def test_random_edge_graph_parity():
+    """-p style graph (extra random edges incl. parallel-edge potential):
+    engine vs oracle on the IDENTICAL graph via from_csr (the -p generator
+    itself is perf-only; parity here is about the Louvain on that input)."""
+    import numpy as np
+    from minivite_amd import Graph, Engine
+    from oracle.oracle import OracleGraph, louvain, sha
+    nv = 65536
+    g = Graph.rgg(nv, 0, 1, random_edge_percent=4.0)
+    xadj, tails, w = g.arrays()
+    og = OracleGraph.from_csr(nv, 1, np.array([0, nv], dtype=np.int64),
+                              [(xadj, tails, w)])
+    omod, oiters, ott, otm = louvain(og, trace=True)
+    og.free()
+    e = Engine(device=0)
+    e.load_graph(g)
+    e.set_trace(64)
+    mod, iters = e.run()
+    tt, tm = e.trace(iters)
+    e.destroy()
+    g.free()
+    assert iters == oiters
+    assert float(mod).hex() == float(omod).hex()
+    for k in range(iters):
+        assert sha(tt[k]) == sha(ott[k]), f"iteration {k+1}"
