@@ -255,6 +255,25 @@ __global__ void k_fill_sell(i64 lnv, const unsigned *__restrict__ perm,
     }
 }
 
+// per-thread spill extents for skewed graphs: with the degree-DESCENDING
+// SELL order, grid-stride thread t's largest vertex is its first position,
+// so its spill need is max(deg_sorted[t] - min_slots, 1); offsets are the
+// host-side prefix sum of these
+__global__ void k_spill_need(i64 nthreads, i64 lnv,
+                             const unsigned *__restrict__ perm,
+                             const unsigned *__restrict__ deg, int min_slots,
+                             i64 *__restrict__ need) {
+    for (i64 t = blockIdx.x * (i64)blockDim.x + threadIdx.x; t < nthreads;
+         t += (i64)gridDim.x * blockDim.x) {
+        i64 n = 1;
+        if (t < lnv) {
+            const i64 d = (i64)deg[perm[t]] - min_slots;
+            n = d > 1 ? d : 1;
+        }
+        need[t] = n;
+    }
+}
+
 // translate a global-id list into internal indices (svdata, once per run)
 __global__ void k_to_internal(i64 n, const i64 *__restrict__ gids, i64 base,
                               const unsigned *__restrict__ sigma_inv,
@@ -393,15 +412,15 @@ __global__ __launch_bounds__(256) void k4_sweep(
     const Info16 *__restrict__ rc_info, Info16 *__restrict__ rcu,
     double constant, i64 *__restrict__ targetComm,
     double *__restrict__ clusterWeight, i64 *__restrict__ spill_keys,
-    double *__restrict__ spill_acc, int spill_max) {
+    double *__restrict__ spill_acc, const i64 *__restrict__ spill_off) {
     extern __shared__ __attribute__((aligned(16))) char smem[];
     i64 *skey = reinterpret_cast<i64 *>(smem);
     double *sacc = reinterpret_cast<double *>(smem + sizeof(i64) * SLOTS * blockDim.x);
     const int tid = threadIdx.x;
     const i64 gthread = blockIdx.x * (i64)blockDim.x + threadIdx.x;
     const i64 stride = (i64)gridDim.x * blockDim.x;
-    i64 *myspill_k = spill_keys + gthread * spill_max;
-    double *myspill_a = spill_acc + gthread * spill_max;
+    i64 *myspill_k = spill_keys + spill_off[gthread];
+    double *myspill_a = spill_acc + spill_off[gthread];
 
     for (i64 s = gthread; s < lnv; s += stride) {
         const i64 i = perm[s];          // internal vertex index
@@ -728,10 +747,13 @@ struct mv_engine {
     void *d_cub_tmp = nullptr;
     size_t cub_tmp_bytes = 0;
 
-    // spill for K4
+    // spill for K4 (per-thread extents; uniform for flat degree
+    // distributions, degree-prefix-sized for skewed ones)
     i64 *d_spill_k = nullptr;
     double *d_spill_a = nullptr;
-    int spill_max = 0;
+    i64 *d_spill_off = nullptr;
+    i64 spill_elems = 0;
+    int skewed = 0;
     int sweep_grid = 0;
 
     // trace
@@ -880,12 +902,15 @@ int mv_engine_load_graph(mv_engine *e, const mv_graph *g) {
     HIP_CHECK(hipMalloc(&e->d_partials, 16 * nblocks));
 
     // K4 geometry: 256-thread blocks, 8 LDS slots/lane (32 KiB/block ->
-    // 4-5 blocks/CU), spill region covers the max degree
+    // 4-5 blocks/CU). Spill sizing: uniform per-thread max_degree extents
+    // for flat distributions; for skewed graphs (max degree > 256) the
+    // SELL order is forced degree-descending and extents follow the
+    // per-thread first-vertex degree (k_spill_need), so the total stays
+    // O(lne) instead of O(threads * max_degree).
     e->sweep_grid = grid_for(lnv, 256, 2048);
-    e->spill_max = (int)std::max<i64>(e->max_degree, 1); // covers any SLOTS
+    e->skewed = e->max_degree > 256;
     const i64 nthreads = (i64)e->sweep_grid * 256;
-    HIP_CHECK(hipMalloc(&e->d_spill_k, 8 * nthreads * e->spill_max));
-    HIP_CHECK(hipMalloc(&e->d_spill_a, 8 * nthreads * e->spill_max));
+    HIP_CHECK(hipMalloc(&e->d_spill_off, 8 * nthreads));
 
     HIP_CHECK(hipDeviceSynchronize());
     e->stats = mv_stats{};
@@ -1039,7 +1064,7 @@ double mv_engine_run(mv_engine *e, double lower, double thresh,
         // otherwise (load balance for skewed degree distributions).
         k_degrees<<<grid_for(lnv), 256, 0, st>>>(lnv, e->d_sigma, e->d_xadj,
                                                  e->d_deg);
-        if (e->has_hint) {
+        if (e->has_hint && !e->skewed) {
             k_iota32<<<grid_for(lnv), 256, 0, st>>>(lnv, e->d_perm);
         } else {
             k_iota32<<<grid_for(lnv), 256, 0, st>>>(lnv, e->d_iota);
@@ -1094,6 +1119,49 @@ double mv_engine_run(mv_engine *e, double lower, double thresh,
                 e->unit_weights ? nullptr : e->d_sell_w);
             HIP_CHECK(hipFree(d_tmp2));
             HIP_CHECK(hipFree(d_sizes));
+        }
+
+        // per-thread spill extents
+        {
+            const i64 nthreads = (i64)e->sweep_grid * 256;
+            i64 total;
+            if (!e->skewed) {
+                const i64 per = std::max<i64>(e->max_degree, 1);
+                std::vector<i64> off(nthreads);
+                for (i64 t = 0; t < nthreads; t++) off[t] = t * per;
+                HIP_CHECK(hipMemcpyAsync(e->d_spill_off, off.data(),
+                                         8 * nthreads, hipMemcpyHostToDevice,
+                                         st));
+                HIP_CHECK(hipStreamSynchronize(st));
+                total = nthreads * per;
+            } else {
+                i64 *d_need = nullptr;
+                HIP_CHECK(hipMalloc(&d_need, 8 * nthreads));
+                k_spill_need<<<grid_for(nthreads), 256, 0, st>>>(
+                    nthreads, lnv, e->d_perm, e->d_deg, 4, d_need);
+                std::vector<i64> need(nthreads), off(nthreads);
+                HIP_CHECK(hipMemcpyAsync(need.data(), d_need, 8 * nthreads,
+                                         hipMemcpyDeviceToHost, st));
+                HIP_CHECK(hipStreamSynchronize(st));
+                i64 acc = 0;
+                for (i64 t = 0; t < nthreads; t++) {
+                    off[t] = acc;
+                    acc += need[t];
+                }
+                total = acc;
+                HIP_CHECK(hipMemcpyAsync(e->d_spill_off, off.data(),
+                                         8 * nthreads, hipMemcpyHostToDevice,
+                                         st));
+                HIP_CHECK(hipStreamSynchronize(st));
+                HIP_CHECK(hipFree(d_need));
+            }
+            if (total > e->spill_elems) {
+                if (e->d_spill_k) HIP_CHECK(hipFree(e->d_spill_k));
+                if (e->d_spill_a) HIP_CHECK(hipFree(e->d_spill_a));
+                e->spill_elems = total;
+                HIP_CHECK(hipMalloc(&e->d_spill_k, 8 * total));
+                HIP_CHECK(hipMalloc(&e->d_spill_a, 8 * total));
+            }
         }
     }
 
@@ -1273,7 +1341,7 @@ double mv_engine_run(mv_engine *e, double lower, double thresh,
                     e->d_ghost_comm, e->d_vdeg, e->d_sigma, e->d_cinfo,
                     e->d_cupd, e->d_rc_ids, nrc, e->d_rc_info, e->d_rcu,
                     constant, d_target, e->d_cw, e->d_spill_k, e->d_spill_a,
-                    e->spill_max);
+                    e->d_spill_off);
         };
         auto dispatch_slots = [&](auto unit_tag) {
             switch (slots) {
