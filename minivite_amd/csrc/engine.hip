@@ -259,15 +259,15 @@ __global__ void k_fill_sell(i64 lnv, const unsigned *__restrict__ perm,
 // SELL order, grid-stride thread t's largest vertex is its first position,
 // so its spill need is max(deg_sorted[t] - min_slots, 1); offsets are the
 // host-side prefix sum of these
-__global__ void k_spill_need(i64 nthreads, i64 lnv,
+__global__ void k_spill_need(i64 nthreads, i64 s_begin, i64 lnv,
                              const unsigned *__restrict__ perm,
                              const unsigned *__restrict__ deg, int min_slots,
                              i64 *__restrict__ need) {
     for (i64 t = blockIdx.x * (i64)blockDim.x + threadIdx.x; t < nthreads;
          t += (i64)gridDim.x * blockDim.x) {
         i64 n = 1;
-        if (t < lnv) {
-            const i64 d = (i64)deg[perm[t]] - min_slots;
+        if (s_begin + t < lnv) {
+            const i64 d = (i64)deg[perm[s_begin + t]] - min_slots;
             n = d > 1 ? d : 1;
         }
         need[t] = n;
@@ -402,7 +402,8 @@ __global__ void k_depermute(i64 lnv, const unsigned *__restrict__ sigma_inv,
 // counter[0], dspl.hpp:312-318).
 template <int SLOTS, bool UNIT>
 __global__ __launch_bounds__(256) void k4_sweep(
-    i64 lnv, i64 base, i64 bound, const unsigned *__restrict__ perm,
+    i64 s_begin, i64 lnv, i64 base, i64 bound,
+    const unsigned *__restrict__ perm,
     const unsigned *__restrict__ deg_int, const i64 *__restrict__ chunk_off,
     const int *__restrict__ sell_tidx, const double *__restrict__ sell_w,
     const i64 *__restrict__ currComm, const i64 *__restrict__ ghost_comm,
@@ -422,7 +423,7 @@ __global__ __launch_bounds__(256) void k4_sweep(
     i64 *myspill_k = spill_keys + spill_off[gthread];
     double *myspill_a = spill_acc + spill_off[gthread];
 
-    for (i64 s = gthread; s < lnv; s += stride) {
+    for (i64 s = s_begin + gthread; s < lnv; s += stride) {
         const i64 i = perm[s];          // internal vertex index
         const int deg = (int)deg_int[i];
         const i64 ebase = chunk_off[s >> 6] + (s & 63);
@@ -579,6 +580,159 @@ __global__ __launch_bounds__(256) void k4_sweep(
     }
 }
 
+// ---- K4 high-degree path (wave-per-vertex) ----
+// For skewed graphs (Orkut-class hubs), one WAVE processes one vertex:
+// lanes stride the CSR row (coalesced 8-B tails), aggregate into a
+// per-vertex open-addressed hash in HBM (L2-resident; keys CASed from -1,
+// weights atomicAdd), then a wave-reduce argmax applies the reference
+// tie-break as a total order on (gain, label) — order-independent, so the
+// reduce tree reproduces dspl.hpp:214-215 exactly. Unit-weight graphs
+// only: their sums are integer-exact under any accumulation order; -w
+// skewed graphs take the serial lane path instead (edge-order bit parity).
+template <bool UNIT>
+__global__ __launch_bounds__(256) void k4_sweep_hi(
+    i64 nhi, i64 lnv, i64 base, i64 bound, const unsigned *__restrict__ perm,
+    const unsigned *__restrict__ deg_int, const unsigned *__restrict__ sigma,
+    const unsigned *__restrict__ sigma_inv, const i64 *__restrict__ xadj,
+    const i64 *__restrict__ tails, const double *__restrict__ ew,
+    const i64 *__restrict__ ghosts, i64 nghost,
+    const i64 *__restrict__ currComm, const i64 *__restrict__ ghost_comm,
+    const double *__restrict__ vDegree, const Cinfo *__restrict__ cinfo,
+    Cinfo *__restrict__ cupd, const i64 *__restrict__ rc_ids, i64 nrc,
+    const Info16 *__restrict__ rc_info, Info16 *__restrict__ rcu,
+    double constant, i64 *__restrict__ targetComm,
+    double *__restrict__ clusterWeight, const i64 *__restrict__ hash_off,
+    i64 *__restrict__ hkeys, double *__restrict__ hacc) {
+    const int lane = threadIdx.x & 63;
+    const int wid = threadIdx.x >> 6;
+    const int wpb = blockDim.x >> 6;
+    for (i64 s = (i64)blockIdx.x * wpb + wid; s < nhi;
+         s += (i64)gridDim.x * wpb) {
+        const i64 i = perm[s];
+        const i64 v = sigma[i];
+        const int deg = (int)deg_int[i];
+        const i64 e0 = xadj[v];
+        const i64 cc = currComm[i];
+        const i64 hoff = hash_off[s];
+        const i64 cap = hash_off[s + 1] - hoff; // power of two
+        const i64 vglobal = v + base;
+        double ccDeg;
+        i64 ccSize;
+        if (cc >= base && cc < bound) {
+            const Cinfo c = cinfo[cc - base];
+            ccDeg = c.degree;
+            ccSize = c.size;
+        } else {
+            const i64 q = dev_bsearch(rc_ids, nrc, cc);
+            ccDeg = rc_info[q].degree;
+            ccSize = rc_info[q].size;
+        }
+        double c0 = 0.0, selfLoop = 0.0;
+        for (int k = lane; k < deg; k += 64) {
+            const i64 tail = tails[e0 + k];
+            const double w = UNIT ? 1.0 : ew[e0 + k];
+            if (tail == vglobal) selfLoop += w;
+            const i64 ti = (tail >= base && tail < bound)
+                               ? (i64)sigma_inv[tail - base]
+                               : lnv + dev_lower_bound(ghosts, nghost, tail);
+            const i64 tcomm = (ti < lnv) ? currComm[ti]
+                                         : ghost_comm[ti - lnv];
+            if (tcomm == cc) { c0 += w; continue; }
+            i64 pos = (i64)(((uint64_t)tcomm * 0x9E3779B97F4A7C15ull) >> 32) &
+                      (cap - 1);
+            for (;;) {
+                const i64 prev = (i64)atomicCAS(
+                    (unsigned long long *)&hkeys[hoff + pos],
+                    (unsigned long long)(-1ll), (unsigned long long)tcomm);
+                if (prev == -1 || prev == tcomm) {
+                    atomicAdd(&hacc[hoff + pos], w);
+                    break;
+                }
+                pos = (pos + 1) & (cap - 1);
+            }
+        }
+        // wave sums (integer-exact for unit weights)
+        for (int off = 32; off > 0; off >>= 1) {
+            c0 += __shfl_down(c0, off, 64);
+            selfLoop += __shfl_down(selfLoop, off, 64);
+        }
+        c0 = __shfl(c0, 0, 64);
+        selfLoop = __shfl(selfLoop, 0, 64);
+        const double vdeg = vDegree[i];
+        const double eix = c0 - selfLoop;
+        const double ax = ccDeg - vdeg;
+        // per-lane best over hash slots, then wave-reduce under the total
+        // order (gain desc, label asc); gain <= 0 maps to the neutral
+        // element so it can never win (dspl.hpp:214-215 semantics)
+        double bg = 0.0;
+        i64 bl = INT64_MAX, bs = 0;
+        for (i64 t = lane; t < cap; t += 64) {
+            const i64 y = hkeys[hoff + t];
+            if (y == -1) continue;
+            const double eiy = hacc[hoff + t];
+            double ay;
+            i64 ysz;
+            if (y >= base && y < bound) {
+                const Cinfo c = cinfo[y - base];
+                ay = c.degree;
+                ysz = c.size;
+            } else {
+                const i64 q = dev_bsearch(rc_ids, nrc, y);
+                ay = rc_info[q].degree;
+                ysz = rc_info[q].size;
+            }
+            const double g =
+                2.0 * (eiy - eix) - 2.0 * vdeg * (ay - ax) * constant;
+            if (g > bg || (g == bg && g != 0.0 && y < bl)) {
+                bg = g;
+                bl = y;
+                bs = ysz;
+            }
+        }
+        for (int off = 32; off > 0; off >>= 1) {
+            const double og = __shfl_down(bg, off, 64);
+            const i64 ol = __shfl_down(bl, off, 64);
+            const i64 os = __shfl_down(bs, off, 64);
+            if (og > bg || (og == bg && ol < bl)) {
+                bg = og;
+                bl = ol;
+                bs = os;
+            }
+        }
+        if (lane == 0) {
+            i64 target = (bl == INT64_MAX) ? cc : bl;
+            if (bs == 1 && ccSize == 1 && target > cc) target = cc; // :224-225
+            if (deg == 0) {
+                clusterWeight[i] = 0;
+                target = cc;
+            } else {
+                clusterWeight[i] = c0;
+            }
+            if (target != cc) {
+                if (cc >= base && cc < bound) {
+                    Cinfo *u = &cupd[cc - base];
+                    atomicAdd(&u->degree, -vdeg);
+                    atomic_add_i64(&u->size, -1);
+                } else {
+                    const i64 q = dev_bsearch(rc_ids, nrc, cc);
+                    atomicAdd(&rcu[q].degree, -vdeg);
+                    atomic_add_i64(&rcu[q].size, -1);
+                }
+                if (target >= base && target < bound) {
+                    Cinfo *u = &cupd[target - base];
+                    atomicAdd(&u->degree, vdeg);
+                    atomic_add_i64(&u->size, 1);
+                } else {
+                    const i64 q = dev_bsearch(rc_ids, nrc, target);
+                    atomicAdd(&rcu[q].degree, vdeg);
+                    atomic_add_i64(&rcu[q].size, 1);
+                }
+            }
+            targetComm[i] = target;
+        }
+    }
+}
+
 // ---- K4 first-iteration specialization ----
 // At iteration 1 currComm is the identity (dspl.hpp:145-147), so per vertex:
 // counter[0] collects exactly the self-loop weight (eix = counter[0] -
@@ -595,7 +749,8 @@ __global__ __launch_bounds__(256) void k4_sweep(
 // (ay-0.0) round exactly like the reference's expressions.
 template <bool UNIT>
 __global__ __launch_bounds__(256) void k4_sweep_iter1(
-    i64 lnv, i64 base, i64 bound, const unsigned *__restrict__ perm,
+    i64 s_begin, i64 lnv, i64 base, i64 bound,
+    const unsigned *__restrict__ perm,
     const unsigned *__restrict__ deg_int, const i64 *__restrict__ chunk_off,
     const int *__restrict__ sell_tidx, const double *__restrict__ sell_w,
     const i64 *__restrict__ currComm, const i64 *__restrict__ ghost_comm,
@@ -606,7 +761,7 @@ __global__ __launch_bounds__(256) void k4_sweep_iter1(
     double *__restrict__ clusterWeight) {
     const i64 gthread = blockIdx.x * (i64)blockDim.x + threadIdx.x;
     const i64 stride = (i64)gridDim.x * blockDim.x;
-    for (i64 s = gthread; s < lnv; s += stride) {
+    for (i64 s = s_begin + gthread; s < lnv; s += stride) {
         const i64 i = perm[s];
         const int deg = (int)deg_int[i];
         const i64 ebase = chunk_off[s >> 6] + (s & 63);
@@ -755,6 +910,13 @@ struct mv_engine {
     i64 spill_elems = 0;
     int skewed = 0;
     int sweep_grid = 0;
+
+    // high-degree (wave-per-vertex) path: first nhi sorted positions
+    i64 nhi = 0;
+    i64 *d_hash_off = nullptr; // nhi+1 slot offsets (per-vertex caps are powers of two)
+    i64 *d_hkeys = nullptr;
+    double *d_hacc = nullptr;
+    i64 hash_total = 0;
 
     // trace
     i64 *trace_target = nullptr;
@@ -1064,6 +1226,7 @@ double mv_engine_run(mv_engine *e, double lower, double thresh,
         // otherwise (load balance for skewed degree distributions).
         k_degrees<<<grid_for(lnv), 256, 0, st>>>(lnv, e->d_sigma, e->d_xadj,
                                                  e->d_deg);
+        e->nhi = 0; // set by the skewed branch below when applicable
         if (e->has_hint && !e->skewed) {
             k_iota32<<<grid_for(lnv), 256, 0, st>>>(lnv, e->d_perm);
         } else {
@@ -1079,7 +1242,40 @@ double mv_engine_run(mv_engine *e, double lower, double thresh,
             hipcub::DeviceRadixSort::SortPairsDescending(
                 d_tmp, tb, e->d_deg, d_degs, e->d_iota, e->d_perm, lnv, 0, 32,
                 st);
+            // high-degree split + per-vertex hash regions (unit weights
+            // only: -w needs the serial edge-order path for bit parity)
+            std::vector<unsigned> sdeg(lnv);
+            HIP_CHECK(hipMemcpyAsync(sdeg.data(), d_degs, 4 * lnv,
+                                     hipMemcpyDeviceToHost, st));
             HIP_CHECK(hipStreamSynchronize(st));
+            constexpr unsigned HI_THRESH = 256;
+            i64 nhi = 0;
+            while (nhi < lnv && sdeg[nhi] > HI_THRESH) nhi++;
+            e->nhi = (e->unit_weights && e->skewed) ? nhi : 0;
+            if (e->nhi > 0) {
+                std::vector<i64> hoff(e->nhi + 1);
+                i64 acc = 0;
+                for (i64 t = 0; t < e->nhi; t++) {
+                    hoff[t] = acc;
+                    i64 cap = 64;
+                    while (cap < 2 * (i64)sdeg[t]) cap <<= 1;
+                    acc += cap;
+                }
+                hoff[e->nhi] = acc;
+                if (!e->d_hash_off)
+                    HIP_CHECK(hipMalloc(&e->d_hash_off, 8 * (e->nhi + 1)));
+                HIP_CHECK(hipMemcpyAsync(e->d_hash_off, hoff.data(),
+                                         8 * (e->nhi + 1),
+                                         hipMemcpyHostToDevice, st));
+                if (acc > e->hash_total) {
+                    if (e->d_hkeys) HIP_CHECK(hipFree(e->d_hkeys));
+                    if (e->d_hacc) HIP_CHECK(hipFree(e->d_hacc));
+                    e->hash_total = acc;
+                    HIP_CHECK(hipMalloc(&e->d_hkeys, 8 * acc));
+                    HIP_CHECK(hipMalloc(&e->d_hacc, 8 * acc));
+                }
+                HIP_CHECK(hipStreamSynchronize(st));
+            }
             HIP_CHECK(hipFree(d_tmp));
             HIP_CHECK(hipFree(d_degs));
         }
@@ -1138,7 +1334,7 @@ double mv_engine_run(mv_engine *e, double lower, double thresh,
                 i64 *d_need = nullptr;
                 HIP_CHECK(hipMalloc(&d_need, 8 * nthreads));
                 k_spill_need<<<grid_for(nthreads), 256, 0, st>>>(
-                    nthreads, lnv, e->d_perm, e->d_deg, 4, d_need);
+                    nthreads, e->nhi, lnv, e->d_perm, e->d_deg, 4, d_need);
                 std::vector<i64> need(nthreads), off(nthreads);
                 HIP_CHECK(hipMemcpyAsync(need.data(), d_need, 8 * nthreads,
                                          hipMemcpyDeviceToHost, st));
@@ -1336,7 +1532,7 @@ double mv_engine_run(mv_engine *e, double lower, double thresh,
             constexpr int S = decltype(slots_tag)::value;
             k4_sweep<S, decltype(unit_tag)::value>
                 <<<e->sweep_grid, 256, S * 256 * 16, st>>>(
-                    lnv, e->base, e->bound, e->d_perm, e->d_deg,
+                    e->nhi, lnv, e->base, e->bound, e->d_perm, e->d_deg,
                     e->d_chunk_off, e->d_sell_tidx, e->d_sell_w, d_curr,
                     e->d_ghost_comm, e->d_vdeg, e->d_sigma, e->d_cinfo,
                     e->d_cupd, e->d_rc_ids, nrc, e->d_rc_info, e->d_rcu,
@@ -1355,11 +1551,23 @@ double mv_engine_run(mv_engine *e, double lower, double thresh,
         auto launch_iter1 = [&](auto unit_tag) {
             k4_sweep_iter1<decltype(unit_tag)::value>
                 <<<e->sweep_grid, 256, 0, st>>>(
-                    lnv, e->base, e->bound, e->d_perm, e->d_deg,
+                    e->nhi, lnv, e->base, e->bound, e->d_perm, e->d_deg,
                     e->d_chunk_off, e->d_sell_tidx, e->d_sell_w, d_curr,
                     e->d_ghost_comm, e->d_vdeg, e->d_cupd, e->d_rc_ids, nrc,
                     e->d_rc_info, e->d_rcu, constant, d_target, e->d_cw);
         };
+        if (e->nhi > 0) { // wave-per-vertex hubs (unit weights only)
+            HIP_CHECK(hipMemsetAsync(e->d_hkeys, 0xFF, 8 * e->hash_total, st));
+            HIP_CHECK(hipMemsetAsync(e->d_hacc, 0, 8 * e->hash_total, st));
+            k4_sweep_hi<true><<<grid_for(e->nhi * 64, 256, 2048), 256, 0,
+                                st>>>(
+                e->nhi, lnv, e->base, e->bound, e->d_perm, e->d_deg,
+                e->d_sigma, e->d_sigma_inv, e->d_xadj, e->d_tails, e->d_ew,
+                e->d_ghosts, e->nghost, d_curr, e->d_ghost_comm, e->d_vdeg,
+                e->d_cinfo, e->d_cupd, e->d_rc_ids, nrc, e->d_rc_info,
+                e->d_rcu, constant, d_target, e->d_cw, e->d_hash_off,
+                e->d_hkeys, e->d_hacc);
+        }
         if (numIters == 1 && e->rows_sorted) {
             if (e->unit_weights)
                 launch_iter1(std::integral_constant<bool, true>{});
