@@ -938,6 +938,8 @@ struct mv_engine {
     }
 };
 
+static void build_sell(mv_engine *e);
+
 extern "C" {
 
 int mv_comm_id(void *id_bytes) {
@@ -1074,6 +1076,7 @@ int mv_engine_load_graph(mv_engine *e, const mv_graph *g) {
     const i64 nthreads = (i64)e->sweep_grid * 256;
     HIP_CHECK(hipMalloc(&e->d_spill_off, 8 * nthreads));
 
+    if (e->nranks == 1) build_sell(e);
     HIP_CHECK(hipDeviceSynchronize());
     e->stats = mv_stats{};
     e->stats.edges_local = lne;
@@ -1126,6 +1129,157 @@ static void exchange_counts(mv_engine *e, const std::vector<i64> &mine,
                              hipMemcpyDeviceToHost, e->stream));
     HIP_CHECK(hipStreamSynchronize(e->stream));
     HIP_CHECK(hipFree(d_all));
+}
+
+// Build the SELL image, internal order, spill extents and the high-degree
+// hash regions. Pure graph-layout preparation (a function of the static
+// graph + the ghost list), analogous to the reference's CSR existing
+// before its timed span; called at load for single-rank graphs and right
+// after ghost discovery inside the span for multi-rank ones (the ghost
+// slots feed the tail translation).
+static void build_sell(mv_engine *e) {
+    hipStream_t st = e->stream;
+    const i64 lnv = e->lnv;
+    // build the SELL image over the internal order (tails translated
+    // inline). perm = identity with a spatial hint, degree-sorted
+    // otherwise (load balance for skewed degree distributions).
+    k_degrees<<<grid_for(lnv), 256, 0, st>>>(lnv, e->d_sigma, e->d_xadj,
+                                             e->d_deg);
+    e->nhi = 0; // set by the skewed branch below when applicable
+    if (e->has_hint && !e->skewed) {
+        k_iota32<<<grid_for(lnv), 256, 0, st>>>(lnv, e->d_perm);
+    } else {
+        k_iota32<<<grid_for(lnv), 256, 0, st>>>(lnv, e->d_iota);
+        unsigned *d_degs = nullptr;
+        HIP_CHECK(hipMalloc(&d_degs, 4 * std::max<i64>(lnv, 1)));
+        size_t tb = 0;
+        hipcub::DeviceRadixSort::SortPairsDescending(
+            nullptr, tb, e->d_deg, d_degs, e->d_iota, e->d_perm, lnv, 0,
+            32, st);
+        void *d_tmp = nullptr;
+        HIP_CHECK(hipMalloc(&d_tmp, std::max<size_t>(tb, 1)));
+        hipcub::DeviceRadixSort::SortPairsDescending(
+            d_tmp, tb, e->d_deg, d_degs, e->d_iota, e->d_perm, lnv, 0, 32,
+            st);
+        // high-degree split + per-vertex hash regions (unit weights
+        // only: -w needs the serial edge-order path for bit parity)
+        std::vector<unsigned> sdeg(lnv);
+        HIP_CHECK(hipMemcpyAsync(sdeg.data(), d_degs, 4 * lnv,
+                                 hipMemcpyDeviceToHost, st));
+        HIP_CHECK(hipStreamSynchronize(st));
+        constexpr unsigned HI_THRESH = 256;
+        i64 nhi = 0;
+        while (nhi < lnv && sdeg[nhi] > HI_THRESH) nhi++;
+        e->nhi = (e->unit_weights && e->skewed) ? nhi : 0;
+        if (e->nhi > 0) {
+            std::vector<i64> hoff(e->nhi + 1);
+            i64 acc = 0;
+            for (i64 t = 0; t < e->nhi; t++) {
+                hoff[t] = acc;
+                i64 cap = 64;
+                while (cap < 2 * (i64)sdeg[t]) cap <<= 1;
+                acc += cap;
+            }
+            hoff[e->nhi] = acc;
+            if (!e->d_hash_off)
+                HIP_CHECK(hipMalloc(&e->d_hash_off, 8 * (e->nhi + 1)));
+            HIP_CHECK(hipMemcpyAsync(e->d_hash_off, hoff.data(),
+                                     8 * (e->nhi + 1),
+                                     hipMemcpyHostToDevice, st));
+            if (acc > e->hash_total) {
+                if (e->d_hkeys) HIP_CHECK(hipFree(e->d_hkeys));
+                if (e->d_hacc) HIP_CHECK(hipFree(e->d_hacc));
+                e->hash_total = acc;
+                HIP_CHECK(hipMalloc(&e->d_hkeys, 8 * acc));
+                HIP_CHECK(hipMalloc(&e->d_hacc, 8 * acc));
+            }
+            HIP_CHECK(hipStreamSynchronize(st));
+        }
+        HIP_CHECK(hipFree(d_tmp));
+        HIP_CHECK(hipFree(d_degs));
+    }
+    {
+        i64 *d_sizes = nullptr;
+        HIP_CHECK(hipMalloc(&d_sizes, 8 * std::max<i64>(e->nchunks, 1)));
+        k_chunk_sizes<<<grid_for(e->nchunks), 256, 0, st>>>(
+            e->nchunks, lnv, e->d_perm, e->d_deg, d_sizes);
+        HIP_CHECK(hipMemsetAsync(e->d_chunk_off, 0, 8, st));
+        size_t tb2 = 0;
+        hipcub::DeviceScan::InclusiveSum(nullptr, tb2, d_sizes,
+                                         e->d_chunk_off + 1, e->nchunks,
+                                         st);
+        void *d_tmp2 = nullptr;
+        HIP_CHECK(hipMalloc(&d_tmp2, std::max<size_t>(tb2, 1)));
+        hipcub::DeviceScan::InclusiveSum(d_tmp2, tb2, d_sizes,
+                                         e->d_chunk_off + 1, e->nchunks,
+                                         st);
+        i64 total = 0;
+        HIP_CHECK(hipMemcpyAsync(&total, e->d_chunk_off + e->nchunks, 8,
+                                 hipMemcpyDeviceToHost, st));
+        HIP_CHECK(hipStreamSynchronize(st));
+        if (total > e->sell_elems) {
+            if (e->d_sell_tidx) HIP_CHECK(hipFree(e->d_sell_tidx));
+            if (e->d_sell_w) HIP_CHECK(hipFree(e->d_sell_w));
+            e->sell_elems = total;
+            HIP_CHECK(hipMalloc(&e->d_sell_tidx,
+                                4 * std::max<i64>(total, 1)));
+            if (!e->unit_weights)
+                HIP_CHECK(hipMalloc(&e->d_sell_w,
+                                    8 * std::max<i64>(total, 1)));
+        }
+        k_fill_sell<<<grid_for(lnv), 256, 0, st>>>(
+            lnv, e->d_perm, e->d_sigma, e->d_sigma_inv, e->d_xadj,
+            e->d_tails, e->d_ew, e->base, e->bound, e->d_ghosts, e->nghost,
+            e->d_chunk_off, e->d_sell_tidx,
+            e->unit_weights ? nullptr : e->d_sell_w);
+        HIP_CHECK(hipFree(d_tmp2));
+        HIP_CHECK(hipFree(d_sizes));
+    }
+
+    // per-thread spill extents
+    {
+        const i64 nthreads = (i64)e->sweep_grid * 256;
+        i64 total;
+        if (!e->skewed) {
+            const i64 per = std::max<i64>(e->max_degree, 1);
+            std::vector<i64> off(nthreads);
+            for (i64 t = 0; t < nthreads; t++) off[t] = t * per;
+            HIP_CHECK(hipMemcpyAsync(e->d_spill_off, off.data(),
+                                     8 * nthreads, hipMemcpyHostToDevice,
+                                     st));
+            HIP_CHECK(hipStreamSynchronize(st));
+            total = nthreads * per;
+        } else {
+            i64 *d_need = nullptr;
+            HIP_CHECK(hipMalloc(&d_need, 8 * nthreads));
+            k_spill_need<<<grid_for(nthreads), 256, 0, st>>>(
+                nthreads, e->nhi, lnv, e->d_perm, e->d_deg, 4, d_need);
+            std::vector<i64> need(nthreads), off(nthreads);
+            HIP_CHECK(hipMemcpyAsync(need.data(), d_need, 8 * nthreads,
+                                     hipMemcpyDeviceToHost, st));
+            HIP_CHECK(hipStreamSynchronize(st));
+            i64 acc = 0;
+            for (i64 t = 0; t < nthreads; t++) {
+                off[t] = acc;
+                acc += need[t];
+            }
+            total = acc;
+            HIP_CHECK(hipMemcpyAsync(e->d_spill_off, off.data(),
+                                     8 * nthreads, hipMemcpyHostToDevice,
+                                     st));
+            HIP_CHECK(hipStreamSynchronize(st));
+            HIP_CHECK(hipFree(d_need));
+        }
+        if (total > e->spill_elems) {
+            if (e->d_spill_k) HIP_CHECK(hipFree(e->d_spill_k));
+            if (e->d_spill_a) HIP_CHECK(hipFree(e->d_spill_a));
+            e->spill_elems = total;
+            HIP_CHECK(hipMalloc(&e->d_spill_k, 8 * total));
+            HIP_CHECK(hipMalloc(&e->d_spill_a, 8 * total));
+        }
+    }
+
+    HIP_CHECK(hipStreamSynchronize(st));
 }
 
 double mv_engine_run(mv_engine *e, double lower, double thresh,
@@ -1221,144 +1375,7 @@ double mv_engine_run(mv_engine *e, double lower, double thresh,
             e->ssz = 0;
         }
 
-        // build the SELL image over the internal order (tails translated
-        // inline). perm = identity with a spatial hint, degree-sorted
-        // otherwise (load balance for skewed degree distributions).
-        k_degrees<<<grid_for(lnv), 256, 0, st>>>(lnv, e->d_sigma, e->d_xadj,
-                                                 e->d_deg);
-        e->nhi = 0; // set by the skewed branch below when applicable
-        if (e->has_hint && !e->skewed) {
-            k_iota32<<<grid_for(lnv), 256, 0, st>>>(lnv, e->d_perm);
-        } else {
-            k_iota32<<<grid_for(lnv), 256, 0, st>>>(lnv, e->d_iota);
-            unsigned *d_degs = nullptr;
-            HIP_CHECK(hipMalloc(&d_degs, 4 * std::max<i64>(lnv, 1)));
-            size_t tb = 0;
-            hipcub::DeviceRadixSort::SortPairsDescending(
-                nullptr, tb, e->d_deg, d_degs, e->d_iota, e->d_perm, lnv, 0,
-                32, st);
-            void *d_tmp = nullptr;
-            HIP_CHECK(hipMalloc(&d_tmp, std::max<size_t>(tb, 1)));
-            hipcub::DeviceRadixSort::SortPairsDescending(
-                d_tmp, tb, e->d_deg, d_degs, e->d_iota, e->d_perm, lnv, 0, 32,
-                st);
-            // high-degree split + per-vertex hash regions (unit weights
-            // only: -w needs the serial edge-order path for bit parity)
-            std::vector<unsigned> sdeg(lnv);
-            HIP_CHECK(hipMemcpyAsync(sdeg.data(), d_degs, 4 * lnv,
-                                     hipMemcpyDeviceToHost, st));
-            HIP_CHECK(hipStreamSynchronize(st));
-            constexpr unsigned HI_THRESH = 256;
-            i64 nhi = 0;
-            while (nhi < lnv && sdeg[nhi] > HI_THRESH) nhi++;
-            e->nhi = (e->unit_weights && e->skewed) ? nhi : 0;
-            if (e->nhi > 0) {
-                std::vector<i64> hoff(e->nhi + 1);
-                i64 acc = 0;
-                for (i64 t = 0; t < e->nhi; t++) {
-                    hoff[t] = acc;
-                    i64 cap = 64;
-                    while (cap < 2 * (i64)sdeg[t]) cap <<= 1;
-                    acc += cap;
-                }
-                hoff[e->nhi] = acc;
-                if (!e->d_hash_off)
-                    HIP_CHECK(hipMalloc(&e->d_hash_off, 8 * (e->nhi + 1)));
-                HIP_CHECK(hipMemcpyAsync(e->d_hash_off, hoff.data(),
-                                         8 * (e->nhi + 1),
-                                         hipMemcpyHostToDevice, st));
-                if (acc > e->hash_total) {
-                    if (e->d_hkeys) HIP_CHECK(hipFree(e->d_hkeys));
-                    if (e->d_hacc) HIP_CHECK(hipFree(e->d_hacc));
-                    e->hash_total = acc;
-                    HIP_CHECK(hipMalloc(&e->d_hkeys, 8 * acc));
-                    HIP_CHECK(hipMalloc(&e->d_hacc, 8 * acc));
-                }
-                HIP_CHECK(hipStreamSynchronize(st));
-            }
-            HIP_CHECK(hipFree(d_tmp));
-            HIP_CHECK(hipFree(d_degs));
-        }
-        {
-            i64 *d_sizes = nullptr;
-            HIP_CHECK(hipMalloc(&d_sizes, 8 * std::max<i64>(e->nchunks, 1)));
-            k_chunk_sizes<<<grid_for(e->nchunks), 256, 0, st>>>(
-                e->nchunks, lnv, e->d_perm, e->d_deg, d_sizes);
-            HIP_CHECK(hipMemsetAsync(e->d_chunk_off, 0, 8, st));
-            size_t tb2 = 0;
-            hipcub::DeviceScan::InclusiveSum(nullptr, tb2, d_sizes,
-                                             e->d_chunk_off + 1, e->nchunks,
-                                             st);
-            void *d_tmp2 = nullptr;
-            HIP_CHECK(hipMalloc(&d_tmp2, std::max<size_t>(tb2, 1)));
-            hipcub::DeviceScan::InclusiveSum(d_tmp2, tb2, d_sizes,
-                                             e->d_chunk_off + 1, e->nchunks,
-                                             st);
-            i64 total = 0;
-            HIP_CHECK(hipMemcpyAsync(&total, e->d_chunk_off + e->nchunks, 8,
-                                     hipMemcpyDeviceToHost, st));
-            HIP_CHECK(hipStreamSynchronize(st));
-            if (total > e->sell_elems) {
-                if (e->d_sell_tidx) HIP_CHECK(hipFree(e->d_sell_tidx));
-                if (e->d_sell_w) HIP_CHECK(hipFree(e->d_sell_w));
-                e->sell_elems = total;
-                HIP_CHECK(hipMalloc(&e->d_sell_tidx,
-                                    4 * std::max<i64>(total, 1)));
-                if (!e->unit_weights)
-                    HIP_CHECK(hipMalloc(&e->d_sell_w,
-                                        8 * std::max<i64>(total, 1)));
-            }
-            k_fill_sell<<<grid_for(lnv), 256, 0, st>>>(
-                lnv, e->d_perm, e->d_sigma, e->d_sigma_inv, e->d_xadj,
-                e->d_tails, e->d_ew, e->base, e->bound, e->d_ghosts, e->nghost,
-                e->d_chunk_off, e->d_sell_tidx,
-                e->unit_weights ? nullptr : e->d_sell_w);
-            HIP_CHECK(hipFree(d_tmp2));
-            HIP_CHECK(hipFree(d_sizes));
-        }
-
-        // per-thread spill extents
-        {
-            const i64 nthreads = (i64)e->sweep_grid * 256;
-            i64 total;
-            if (!e->skewed) {
-                const i64 per = std::max<i64>(e->max_degree, 1);
-                std::vector<i64> off(nthreads);
-                for (i64 t = 0; t < nthreads; t++) off[t] = t * per;
-                HIP_CHECK(hipMemcpyAsync(e->d_spill_off, off.data(),
-                                         8 * nthreads, hipMemcpyHostToDevice,
-                                         st));
-                HIP_CHECK(hipStreamSynchronize(st));
-                total = nthreads * per;
-            } else {
-                i64 *d_need = nullptr;
-                HIP_CHECK(hipMalloc(&d_need, 8 * nthreads));
-                k_spill_need<<<grid_for(nthreads), 256, 0, st>>>(
-                    nthreads, e->nhi, lnv, e->d_perm, e->d_deg, 4, d_need);
-                std::vector<i64> need(nthreads), off(nthreads);
-                HIP_CHECK(hipMemcpyAsync(need.data(), d_need, 8 * nthreads,
-                                         hipMemcpyDeviceToHost, st));
-                HIP_CHECK(hipStreamSynchronize(st));
-                i64 acc = 0;
-                for (i64 t = 0; t < nthreads; t++) {
-                    off[t] = acc;
-                    acc += need[t];
-                }
-                total = acc;
-                HIP_CHECK(hipMemcpyAsync(e->d_spill_off, off.data(),
-                                         8 * nthreads, hipMemcpyHostToDevice,
-                                         st));
-                HIP_CHECK(hipStreamSynchronize(st));
-                HIP_CHECK(hipFree(d_need));
-            }
-            if (total > e->spill_elems) {
-                if (e->d_spill_k) HIP_CHECK(hipFree(e->d_spill_k));
-                if (e->d_spill_a) HIP_CHECK(hipFree(e->d_spill_a));
-                e->spill_elems = total;
-                HIP_CHECK(hipMalloc(&e->d_spill_k, 8 * total));
-                HIP_CHECK(hipMalloc(&e->d_spill_a, 8 * total));
-            }
-        }
+        if (p > 1) build_sell(e); // p==1: built at load
     }
 
     // ---- distInitLouvain (dspl.hpp:151-172) ----
