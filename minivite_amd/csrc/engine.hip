@@ -90,6 +90,26 @@ struct Info16 {    // wire record for cinfo replies / delta routing: the
     double degree; // id, which both ends know by position
 };
 
+// Community HANDLES: community values inside the engine are
+// (label << 32) | slot, where label is the reference's community id
+// (a global vertex id) and slot is the owner-local INTERNAL index of the
+// label's home vertex (valid only for locally-owned labels; remote
+// handles carry slot 0 and are resolved through the sorted rc_ids
+// label array). The label occupies the high bits, so handle comparison
+// and equality reproduce the reference's label comparison exactly
+// (dspl.hpp:214-215 tie-break included); localCinfo/localCupdate live in
+// INTERNAL order, making candidate-info gathers as spatially local as the
+// community structure itself. Labels on the wire stay plain (handles are
+// never exchanged). Requires nv < 2^31 (checked at load).
+__device__ __forceinline__ i64 h_label(i64 h) { return h >> 32; }
+// Speculation-safe index clamps: the branches below ARE semantically
+// exclusive, but the compiler may if-convert them into two unconditional
+// loads + a value select, so the un-taken side's address must still be
+// dereferenceable (ghost/rc buffers are never null — load_graph keeps
+// 1-element dummies — and indices are clamped non-negative).
+__device__ __forceinline__ i64 clamp0(i64 x) { return x > 0 ? x : 0; }
+__device__ __forceinline__ i64 h_slot(i64 h) { return h & 0xFFFFFFFFll; }
+
 __device__ __forceinline__ i64 dev_lower_bound(const i64 *a, i64 n, i64 key) {
     i64 lo = 0, hi = n;
     while (lo < hi) {
@@ -110,8 +130,8 @@ __device__ __forceinline__ void atomic_add_i64(i64 *p, i64 v) {
               static_cast<unsigned long long>(v));
 }
 
-// ---- K1: vDegree + localCinfo init (dspl.hpp:82-107). vDegree is
-// internal-ordered; cinfo is label-ordered. ----
+// ---- K1: vDegree + localCinfo init (dspl.hpp:82-107); both
+// internal-ordered ----
 __global__ void k1_vertex_degree(i64 lnv, const unsigned *__restrict__ sigma,
                                  const i64 *__restrict__ xadj,
                                  const double *__restrict__ ew, int unit,
@@ -129,7 +149,7 @@ __global__ void k1_vertex_degree(i64 lnv, const unsigned *__restrict__ sigma,
             for (i64 e = e0; e < e1; e++) tw += ew[e]; // sequential edge order
         }
         vDegree[k] = tw;
-        cinfo[v] = {1, tw}; // dspl.hpp:104-105
+        cinfo[k] = {1, tw}; // dspl.hpp:104-105 (internal slot k)
     }
 }
 
@@ -169,9 +189,9 @@ __global__ void k3_init_comm(i64 lnv, i64 base,
                              i64 *__restrict__ curr, i64 *__restrict__ past) {
     for (i64 k = blockIdx.x * (i64)blockDim.x + threadIdx.x; k < lnv;
          k += (i64)gridDim.x * blockDim.x) {
-        const i64 label = (i64)sigma[k] + base;
-        curr[k] = label;
-        past[k] = label;
+        const i64 h = (((i64)sigma[k] + base) << 32) | k; // handle
+        curr[k] = h;
+        past[k] = h;
     }
 }
 
@@ -283,22 +303,36 @@ __global__ void k_to_internal(i64 n, const i64 *__restrict__ gids, i64 base,
         out[k] = sigma_inv[gids[k] - base];
 }
 
+// ghost communities arrive as LABELS; convert to handles
+__global__ void k_labels_to_handles(i64 n, i64 base, i64 bound,
+                                    const unsigned *__restrict__ sigma_inv,
+                                    i64 *__restrict__ vals) {
+    for (i64 k = blockIdx.x * (i64)blockDim.x + threadIdx.x; k < n;
+         k += (i64)gridDim.x * blockDim.x) {
+        const i64 c = vals[k];
+        vals[k] = (c >= base && c < bound)
+                      ? (c << 32) | (i64)sigma_inv[c - base]
+                      : c << 32;
+    }
+}
+
 // ---- K8: scdata gather (dspl.hpp:559-571) ----
 __global__ void k8_gather_comms(i64 n, const unsigned *__restrict__ svdata_int,
                                 const i64 *__restrict__ currComm,
                                 i64 *__restrict__ out) {
     for (i64 k = blockIdx.x * (i64)blockDim.x + threadIdx.x; k < n;
          k += (i64)gridDim.x * blockDim.x)
-        out[k] = currComm[svdata_int[k]];
+        out[k] = currComm[svdata_int[k]] >> 32; // labels on the wire
 }
 
 // ---- candidate remote communities (dspl.hpp:670-700) ----
-__global__ void k_filter_remote(i64 n, const i64 *__restrict__ vals, i64 base,
-                                i64 bound, i64 *__restrict__ out,
+__global__ void k_filter_remote(i64 n, const i64 *__restrict__ vals,
+                                int shift, i64 base, i64 bound,
+                                i64 *__restrict__ out,
                                 unsigned long long *__restrict__ count) {
     for (i64 k = blockIdx.x * (i64)blockDim.x + threadIdx.x; k < n;
          k += (i64)gridDim.x * blockDim.x) {
-        const i64 c = vals[k];
+        const i64 c = vals[k] >> shift; // label
         if (c < base || c >= bound) out[atomicAdd(count, 1ull)] = c;
     }
 }
@@ -313,11 +347,12 @@ __global__ void k_owner_bounds(const i64 *__restrict__ sorted, i64 n,
 
 // ---- K9: cinfo reply gather (dspl.hpp:776-929) ----
 __global__ void k9_reply_info(i64 n, const i64 *__restrict__ req_ids, i64 base,
+                              const unsigned *__restrict__ sigma_inv,
                               const Cinfo *__restrict__ cinfo,
                               Info16 *__restrict__ out) {
     for (i64 k = blockIdx.x * (i64)blockDim.x + threadIdx.x; k < n;
          k += (i64)gridDim.x * blockDim.x) {
-        const Cinfo c = cinfo[req_ids[k] - base];
+        const Cinfo c = cinfo[sigma_inv[req_ids[k] - base]];
         out[k] = {c.size, c.degree};
     }
 }
@@ -325,11 +360,12 @@ __global__ void k9_reply_info(i64 n, const i64 *__restrict__ req_ids, i64 base,
 // apply received deltas to owned cinfo (dspl.hpp:1089-1102; atomics because
 // several senders may address one community)
 __global__ void k_apply_deltas(i64 n, const i64 *__restrict__ ids, i64 base,
+                               const unsigned *__restrict__ sigma_inv,
                                const Info16 *__restrict__ deltas,
                                Cinfo *__restrict__ cinfo) {
     for (i64 k = blockIdx.x * (i64)blockDim.x + threadIdx.x; k < n;
          k += (i64)gridDim.x * blockDim.x) {
-        Cinfo *c = &cinfo[ids[k] - base];
+        Cinfo *c = &cinfo[sigma_inv[ids[k] - base]];
         atomic_add_i64(&c->size, deltas[k].size);
         atomicAdd(&c->degree, deltas[k].degree);
     }
@@ -389,7 +425,7 @@ __global__ void k_depermute(i64 lnv, const unsigned *__restrict__ sigma_inv,
                             i64 *__restrict__ out) {
     for (i64 v = blockIdx.x * (i64)blockDim.x + threadIdx.x; v < lnv;
          v += (i64)gridDim.x * blockDim.x)
-        out[v] = in[sigma_inv[v]];
+        out[v] = in[sigma_inv[v]] >> 32; // handle -> label
 }
 
 // ---- K4: THE sweep (distExecuteLouvainIteration, dspl.hpp:276-405) ----
@@ -413,7 +449,8 @@ __global__ __launch_bounds__(256) void k4_sweep(
     const Info16 *__restrict__ rc_info, Info16 *__restrict__ rcu,
     double constant, i64 *__restrict__ targetComm,
     double *__restrict__ clusterWeight, i64 *__restrict__ spill_keys,
-    double *__restrict__ spill_acc, const i64 *__restrict__ spill_off) {
+    double *__restrict__ spill_acc, const i64 *__restrict__ spill_off,
+    int dbg_skip) {
     extern __shared__ __attribute__((aligned(16))) char smem[];
     i64 *skey = reinterpret_cast<i64 *>(smem);
     double *sacc = reinterpret_cast<double *>(smem + sizeof(i64) * SLOTS * blockDim.x);
@@ -427,15 +464,17 @@ __global__ __launch_bounds__(256) void k4_sweep(
         const i64 i = perm[s];          // internal vertex index
         const int deg = (int)deg_int[i];
         const i64 ebase = chunk_off[s >> 6] + (s & 63);
-        const i64 cc = currComm[i];
+        const i64 cc = currComm[i]; // handle
+        const i64 cl = h_label(cc);
+        const bool cc_local = (cl >= base && cl < bound);
         double ccDeg;
         i64 ccSize;
-        if (cc >= base && cc < bound) { // dspl.hpp:296-307
-            const Cinfo c = cinfo[cc - base];
+        if (cc_local) { // dspl.hpp:296-307
+            const Cinfo c = cinfo[h_slot(cc)];
             ccDeg = c.degree;
             ccSize = c.size;
         } else {
-            const i64 q = dev_bsearch(rc_ids, nrc, cc);
+            const i64 q = clamp0(dev_bsearch(rc_ids, nrc, cl));
             ccDeg = rc_info[q].degree;
             ccSize = rc_info[q].size;
         }
@@ -467,7 +506,7 @@ __global__ __launch_bounds__(256) void k4_sweep(
 #pragma unroll
                 for (int j = 0; j < CH; j++)
                     cb[j] = (tb[j] < lnv) ? currComm[tb[j]]
-                                          : ghost_comm[tb[j] - lnv];
+                                          : ghost_comm[clamp0(tb[j] - lnv)];
                 for (int j = 0; j < m; j++) {
                     const i64 tidx = tb[j];
                     const double w = UNIT ? 1.0 : wb[j];
@@ -489,6 +528,7 @@ __global__ __launch_bounds__(256) void k4_sweep(
                         ns++;
                         continue;
                     }
+                    if (dbg_skip & 4) continue; // DEBUG: drop spill
                     for (int t = 0; t < nspill; t++) {
                         if (myspill_k[t] == tcomm) {
                             myspill_a[t] += w;
@@ -505,49 +545,39 @@ __global__ __launch_bounds__(256) void k4_sweep(
             }
             clusterWeight[i] = c0; // dspl.hpp:318 onto the K5-zeroed value
 
-            // distGetMaxIndex (dspl.hpp:174-228); candidate info gathers
-            // batched 8-wide like the edge loop (independent 16-B loads)
+            // distGetMaxIndex (dspl.hpp:174-228)
             const double vdeg = vDegree[i];
             const double eix = c0 - selfLoop;
             const double ax = ccDeg - vdeg;
             double maxGain = 0.0;
             i64 maxIndex = cc, maxSize = ccSize;
-            const int tot = ns + nspill;
-            for (int t0 = 0; t0 < tot; t0 += CH) {
-                const int m2 = min(CH, tot - t0);
-                i64 yb[CH];
-                double eb[CH], ab[CH];
-                i64 zb[CH];
-#pragma unroll
-                for (int j = 0; j < CH; j++) {
-                    const int t = (j < m2) ? t0 + j : t0;
-                    yb[j] = (t < ns) ? skey[t * blockDim.x + tid]
-                                     : myspill_k[t - ns];
-                    eb[j] = (t < ns) ? sacc[t * blockDim.x + tid]
-                                     : myspill_a[t - ns];
+            const int tot = (dbg_skip & 2) ? 0 : ns + nspill;
+            for (int t = 0; t < tot; t++) {
+                const i64 y = (t < ns) ? skey[t * blockDim.x + tid]
+                                       : myspill_k[t - ns];
+                const double eiy = (t < ns) ? sacc[t * blockDim.x + tid]
+                                            : myspill_a[t - ns];
+                if (dbg_skip & 16) { continue; } // DEBUG: read-only scan
+                const i64 yl = h_label(y);
+                double ay;
+                i64 ysz;
+                if (dbg_skip & 8) { ay = 0.0; ysz = 1; } // DEBUG: no lookup
+                else if (yl >= base && yl < bound) {
+                    const Cinfo c = cinfo[h_slot(y)];
+                    ay = c.degree;
+                    ysz = c.size;
+                } else {
+                    const i64 q = clamp0(dev_bsearch(rc_ids, nrc, yl));
+                    ay = rc_info[q].degree;
+                    ysz = rc_info[q].size;
                 }
-#pragma unroll
-                for (int j = 0; j < CH; j++) {
-                    if (yb[j] >= base && yb[j] < bound) {
-                        const Cinfo c = cinfo[yb[j] - base];
-                        ab[j] = c.degree;
-                        zb[j] = c.size;
-                    } else {
-                        const i64 q = dev_bsearch(rc_ids, nrc, yb[j]);
-                        ab[j] = rc_info[q].degree;
-                        zb[j] = rc_info[q].size;
-                    }
-                }
-                for (int j = 0; j < m2; j++) {
-                    const double curGain = 2.0 * (eb[j] - eix) -
-                                           2.0 * vdeg * (ab[j] - ax) * constant; // :212
-                    if (curGain > maxGain ||
-                        (curGain == maxGain && curGain != 0.0 &&
-                         yb[j] < maxIndex)) {
-                        maxGain = curGain;
-                        maxIndex = yb[j];
-                        maxSize = zb[j];
-                    }
+                const double curGain =
+                    2.0 * (eiy - eix) - 2.0 * vdeg * (ay - ax) * constant; // :212
+                if (curGain > maxGain ||
+                    (curGain == maxGain && curGain != 0.0 && y < maxIndex)) {
+                    maxGain = curGain;
+                    maxIndex = y;
+                    maxSize = ysz;
                 }
             }
             if (maxSize == 1 && ccSize == 1 && maxIndex > cc) // :224-225
@@ -555,28 +585,29 @@ __global__ __launch_bounds__(256) void k4_sweep(
             target = maxIndex;
         }
 
-        if (target != cc) { // 4-case updates (dspl.hpp:331-399)
+        if (target != cc && !(dbg_skip & 1)) { // 4-case updates (dspl.hpp:331-399)
             const double vdeg = vDegree[i];
-            if (cc >= base && cc < bound) {
-                Cinfo *u = &cupd[cc - base];
+            if (cc_local) {
+                Cinfo *u = &cupd[h_slot(cc)];
                 atomicAdd(&u->degree, -vdeg);
                 atomic_add_i64(&u->size, -1);
             } else {
-                const i64 q = dev_bsearch(rc_ids, nrc, cc);
+                const i64 q = dev_bsearch(rc_ids, nrc, cl);
                 atomicAdd(&rcu[q].degree, -vdeg);
                 atomic_add_i64(&rcu[q].size, -1);
             }
-            if (target >= base && target < bound) {
-                Cinfo *u = &cupd[target - base];
+            const i64 tl = h_label(target);
+            if (tl >= base && tl < bound) {
+                Cinfo *u = &cupd[h_slot(target)];
                 atomicAdd(&u->degree, vdeg);
                 atomic_add_i64(&u->size, 1);
             } else {
-                const i64 q = dev_bsearch(rc_ids, nrc, target);
+                const i64 q = dev_bsearch(rc_ids, nrc, tl);
                 atomicAdd(&rcu[q].degree, vdeg);
                 atomic_add_i64(&rcu[q].size, 1);
             }
         }
-        targetComm[i] = target; // dspl.hpp:404
+        targetComm[i] = target; // dspl.hpp:404 (handle)
     }
 }
 
@@ -612,18 +643,20 @@ __global__ __launch_bounds__(256) void k4_sweep_hi(
         const i64 v = sigma[i];
         const int deg = (int)deg_int[i];
         const i64 e0 = xadj[v];
-        const i64 cc = currComm[i];
+        const i64 cc = currComm[i]; // handle
+        const i64 cl = h_label(cc);
+        const bool cc_local = (cl >= base && cl < bound);
         const i64 hoff = hash_off[s];
         const i64 cap = hash_off[s + 1] - hoff; // power of two
         const i64 vglobal = v + base;
         double ccDeg;
         i64 ccSize;
-        if (cc >= base && cc < bound) {
-            const Cinfo c = cinfo[cc - base];
+        if (cc_local) {
+            const Cinfo c = cinfo[h_slot(cc)];
             ccDeg = c.degree;
             ccSize = c.size;
         } else {
-            const i64 q = dev_bsearch(rc_ids, nrc, cc);
+            const i64 q = clamp0(dev_bsearch(rc_ids, nrc, cl));
             ccDeg = rc_info[q].degree;
             ccSize = rc_info[q].size;
         }
@@ -670,14 +703,15 @@ __global__ __launch_bounds__(256) void k4_sweep_hi(
             const i64 y = hkeys[hoff + t];
             if (y == -1) continue;
             const double eiy = hacc[hoff + t];
+            const i64 yl = h_label(y);
             double ay;
             i64 ysz;
-            if (y >= base && y < bound) {
-                const Cinfo c = cinfo[y - base];
+            if (yl >= base && yl < bound) {
+                const Cinfo c = cinfo[h_slot(y)];
                 ay = c.degree;
                 ysz = c.size;
             } else {
-                const i64 q = dev_bsearch(rc_ids, nrc, y);
+                const i64 q = clamp0(dev_bsearch(rc_ids, nrc, yl));
                 ay = rc_info[q].degree;
                 ysz = rc_info[q].size;
             }
@@ -709,21 +743,22 @@ __global__ __launch_bounds__(256) void k4_sweep_hi(
                 clusterWeight[i] = c0;
             }
             if (target != cc) {
-                if (cc >= base && cc < bound) {
-                    Cinfo *u = &cupd[cc - base];
+                if (cc_local) {
+                    Cinfo *u = &cupd[h_slot(cc)];
                     atomicAdd(&u->degree, -vdeg);
                     atomic_add_i64(&u->size, -1);
                 } else {
-                    const i64 q = dev_bsearch(rc_ids, nrc, cc);
+                    const i64 q = dev_bsearch(rc_ids, nrc, cl);
                     atomicAdd(&rcu[q].degree, -vdeg);
                     atomic_add_i64(&rcu[q].size, -1);
                 }
-                if (target >= base && target < bound) {
-                    Cinfo *u = &cupd[target - base];
+                const i64 tl2 = h_label(target);
+                if (tl2 >= base && tl2 < bound) {
+                    Cinfo *u = &cupd[h_slot(target)];
                     atomicAdd(&u->degree, vdeg);
                     atomic_add_i64(&u->size, 1);
                 } else {
-                    const i64 q = dev_bsearch(rc_ids, nrc, target);
+                    const i64 q = dev_bsearch(rc_ids, nrc, tl2);
                     atomicAdd(&rcu[q].degree, vdeg);
                     atomic_add_i64(&rcu[q].size, 1);
                 }
@@ -788,7 +823,7 @@ __global__ __launch_bounds__(256) void k4_sweep_iter1(
 #pragma unroll
             for (int j = 0; j < CH; j++) {
                 cb[j] = (tb[j] < lnv) ? currComm[tb[j]]
-                                      : ghost_comm[tb[j] - lnv];
+                                      : ghost_comm[clamp0(tb[j] - lnv)];
                 vb[j] = (tb[j] < lnv) ? vDegree[tb[j]] : 0.0;
             }
             for (int j = 0; j < m; j++) {
@@ -802,11 +837,12 @@ __global__ __launch_bounds__(256) void k4_sweep_iter1(
                     if (g > maxGain) { maxGain = g; maxIndex = pend_label; }
                 }
                 prev = tidx;
-                pend_label = cb[j];
+                pend_label = cb[j]; // handle (label-ordered like the tails)
                 eiy = w;
                 pend_ay = (tidx < lnv)
                               ? vb[j]
-                              : rc_info[dev_bsearch(rc_ids, nrc, cb[j])].degree;
+                              : rc_info[clamp0(dev_bsearch(
+                                    rc_ids, nrc, h_label(cb[j])))].degree;
                 pend = true;
             }
         }
@@ -814,23 +850,25 @@ __global__ __launch_bounds__(256) void k4_sweep_iter1(
             const double g = 2.0 * eiy - 2.0 * vdeg * pend_ay * constant;
             if (g > maxGain) { maxGain = g; maxIndex = pend_label; }
         }
-        if (maxIndex > cc) maxIndex = cc; // singleton guard: all sizes are 1
+        if (maxIndex > cc) maxIndex = cc; // singleton guard (handle order
+                                          // == label order)
         clusterWeight[i] = c0;            // dspl.hpp:318 (eix == 0)
         if (maxIndex != cc) {             // cc is local at iteration 1
-            Cinfo *u = &cupd[cc - base];
+            Cinfo *u = &cupd[h_slot(cc)];
             atomicAdd(&u->degree, -vdeg);
             atomic_add_i64(&u->size, -1);
-            if (maxIndex >= base && maxIndex < bound) {
-                Cinfo *t = &cupd[maxIndex - base];
+            const i64 tl = h_label(maxIndex);
+            if (tl >= base && tl < bound) {
+                Cinfo *t = &cupd[h_slot(maxIndex)];
                 atomicAdd(&t->degree, vdeg);
                 atomic_add_i64(&t->size, 1);
             } else {
-                const i64 q = dev_bsearch(rc_ids, nrc, maxIndex);
+                const i64 q = dev_bsearch(rc_ids, nrc, tl);
                 atomicAdd(&rcu[q].degree, vdeg);
                 atomic_add_i64(&rcu[q].size, 1);
             }
         }
-        targetComm[i] = maxIndex;
+        targetComm[i] = maxIndex; // handle
     }
 }
 
@@ -1001,6 +1039,10 @@ void mv_engine_destroy(mv_engine *e) {
 int mv_engine_load_graph(mv_engine *e, const mv_graph *g) {
     HIP_CHECK(hipSetDevice(e->device));
     e->nv = mv_graph_nv(g);
+    if (e->nv >= (1ll << 31)) { // community-handle packing needs 31-bit ids
+        std::fprintf(stderr, "mv_engine_load_graph: nv >= 2^31 unsupported\n");
+        return -1;
+    }
     e->lnv = mv_graph_lnv(g);
     e->lne = mv_graph_lne(g);
     e->parts_h.assign(mv_graph_parts(g), mv_graph_parts(g) + e->nranks + 1);
@@ -1059,6 +1101,13 @@ int mv_engine_load_graph(mv_engine *e, const mv_graph *g) {
     HIP_CHECK(hipMalloc(&e->d_cw, 8 * lnv));
     HIP_CHECK(hipMalloc(&e->d_cinfo, sizeof(Cinfo) * lnv));
     HIP_CHECK(hipMalloc(&e->d_cupd, sizeof(Cinfo) * lnv));
+    // never-null dummies for pointers that stay unused at nranks==1 but
+    // may still be address-computed by if-converted loads
+    if (!e->d_ghost_comm)
+        HIP_CHECK(hipMalloc(&e->d_ghost_comm, 16));
+    if (!e->d_rc_ids) HIP_CHECK(hipMalloc(&e->d_rc_ids, 16));
+    if (!e->d_rc_info) HIP_CHECK(hipMalloc(&e->d_rc_info, sizeof(Info16)));
+    if (!e->d_rcu) HIP_CHECK(hipMalloc(&e->d_rcu, sizeof(Info16)));
     HIP_CHECK(hipMalloc(&e->d_count, 8));
     HIP_CHECK(hipMalloc(&e->d_bounds, 8 * (e->nranks + 1)));
     HIP_CHECK(hipMalloc(&e->d_red, 16));
@@ -1282,6 +1331,15 @@ static void build_sell(mv_engine *e) {
     HIP_CHECK(hipStreamSynchronize(st));
 }
 
+#define PHASE(tag)                                                            \
+    do {                                                                      \
+        if (getenv("MV_PHASE_DEBUG")) {                                       \
+            HIP_CHECK(hipStreamSynchronize(e->stream));                       \
+            std::fprintf(stderr, "[phase] %s\n", tag);                        \
+            std::fflush(stderr);                                              \
+        }                                                                     \
+    } while (0)
+
 double mv_engine_run(mv_engine *e, double lower, double thresh,
                      int *iters_out) {
     HIP_CHECK(hipSetDevice(e->device));
@@ -1378,10 +1436,12 @@ double mv_engine_run(mv_engine *e, double lower, double thresh,
         if (p > 1) build_sell(e); // p==1: built at load
     }
 
+    PHASE("setup-done");
     // ---- distInitLouvain (dspl.hpp:151-172) ----
     k1_vertex_degree<<<grid_for(lnv), 256, 0, st>>>(
         lnv, e->d_sigma, e->d_xadj, e->d_ew, e->unit_weights, e->d_vdeg,
         e->d_cinfo);
+    PHASE("k1-done");
     const int nblocks = grid_for(lnv);
     {
         auto f = [vd = e->d_vdeg] __device__(i64 i, double &a, double &b) {
@@ -1405,9 +1465,59 @@ double mv_engine_run(mv_engine *e, double lower, double thresh,
         HIP_CHECK(hipStreamSynchronize(st));
     }
     const double constant = 1.0 / totalW; // dspl.hpp:129
+    PHASE("k2-done");
     k3_init_comm<<<grid_for(lnv), 256, 0, st>>>(lnv, e->base, e->d_sigma,
                                                 e->d_curr, e->d_past);
     HIP_CHECK(hipStreamSynchronize(st));
+    PHASE("k3-done");
+    if (getenv("MV_VERIFY")) {
+        HIP_CHECK(hipStreamSynchronize(st));
+        std::vector<i64> hc(lnv);
+        HIP_CHECK(hipMemcpy(hc.data(), e->d_curr, 8 * lnv,
+                            hipMemcpyDeviceToHost));
+        std::vector<unsigned> hs(lnv);
+        HIP_CHECK(hipMemcpy(hs.data(), e->d_sigma, 4 * lnv,
+                            hipMemcpyDeviceToHost));
+        i64 bad = 0;
+        for (i64 k = 0; k < lnv; k++) {
+            const i64 lab = hc[k] >> 32, sl = hc[k] & 0xFFFFFFFFll;
+            if (lab != (i64)hs[k] + e->base || sl != k)
+                if (bad++ < 3)
+                    std::fprintf(stderr, "VERIFY curr[%lld]=%llx exp label "
+                                 "%lld\n", (long long)k,
+                                 (unsigned long long)hc[k],
+                                 (long long)((i64)hs[k] + e->base));
+        }
+        std::fprintf(stderr, "VERIFY curr: %lld bad of %lld\n",
+                     (long long)bad, (long long)lnv);
+        // SELL tidx bounds
+        std::vector<int> st_h(e->sell_elems);
+        HIP_CHECK(hipMemcpy(st_h.data(), e->d_sell_tidx, 4 * e->sell_elems,
+                            hipMemcpyDeviceToHost));
+        std::vector<i64> co(e->nchunks + 1);
+        HIP_CHECK(hipMemcpy(co.data(), e->d_chunk_off, 8 * (e->nchunks + 1),
+                            hipMemcpyDeviceToHost));
+        std::vector<unsigned> dg(lnv), pm(lnv);
+        HIP_CHECK(hipMemcpy(dg.data(), e->d_deg, 4 * lnv,
+                            hipMemcpyDeviceToHost));
+        HIP_CHECK(hipMemcpy(pm.data(), e->d_perm, 4 * lnv,
+                            hipMemcpyDeviceToHost));
+        bad = 0;
+        for (i64 sp = 0; sp < lnv; sp++) {
+            const i64 iint = pm[sp];
+            const i64 eb2 = co[sp >> 6] + (sp & 63);
+            for (unsigned k = 0; k < dg[iint]; k++) {
+                const int ti = st_h[eb2 + (i64)k * 64];
+                if (ti < 0 || ti >= lnv + e->nghost)
+                    if (bad++ < 3)
+                        std::fprintf(stderr, "VERIFY sell s=%lld k=%u ti=%d\n",
+                                     (long long)sp, k, ti);
+            }
+        }
+        std::fprintf(stderr, "VERIFY sell: %lld bad (elems %lld)\n",
+                     (long long)bad, (long long)e->sell_elems);
+        std::fflush(stderr);
+    }
     e->stats.setup_ms =
         std::chrono::duration<double, std::milli>(
             std::chrono::steady_clock::now() - t_setup0)
@@ -1460,10 +1570,11 @@ double mv_engine_run(mv_engine *e, double lower, double thresh,
             }
             HIP_CHECK(hipMemsetAsync(e->d_count, 0, 8, st));
             k_filter_remote<<<grid_for(std::max<i64>(e->nghost, 1)), 256, 0,
-                              st>>>(e->nghost, e->d_ghost_comm, e->base,
-                                    e->bound, e->d_cand, e->d_count);
+                              st>>>(e->nghost, e->d_ghost_comm, /*shift*/ 0,
+                                    e->base, e->bound, e->d_cand, e->d_count);
             k_filter_remote<<<grid_for(lnv), 256, 0, st>>>(
-                lnv, d_curr, e->base, e->bound, e->d_cand, e->d_count);
+                lnv, d_curr, /*shift*/ 32, e->base, e->bound, e->d_cand,
+                e->d_count);
             unsigned long long ncand = 0;
             HIP_CHECK(hipMemcpyAsync(&ncand, e->d_count, 8,
                                      hipMemcpyDeviceToHost, st));
@@ -1477,6 +1588,10 @@ double mv_engine_run(mv_engine *e, double lower, double thresh,
             hipcub::DeviceSelect::Unique(e->d_cub_tmp, tb, e->d_cand_sorted,
                                          e->d_rc_ids, d_nrc, (int64_t)ncand, st);
             HIP_CHECK(hipMemcpyAsync(&nrc, d_nrc, 8, hipMemcpyDeviceToHost, st));
+            // ghost communities: labels -> handles for the sweep
+            k_labels_to_handles<<<grid_for(std::max<i64>(e->nghost, 1)), 256,
+                                  0, st>>>(e->nghost, e->base, e->bound,
+                                           e->d_sigma_inv, e->d_ghost_comm);
             HIP_CHECK(hipStreamSynchronize(st));
 
             // ---- halo #1b/#1c/#1d: request (size,degree) of those
@@ -1506,7 +1621,8 @@ double mv_engine_run(mv_engine *e, double lower, double thresh,
             rccl_alltoallv(e, e->d_rc_ids, rc_bounds.data(), e->d_req_ids,
                            req_off.data(), 8, ncclInt64, 8);
             k9_reply_info<<<grid_for(std::max<i64>(nreq, 1)), 256, 0, st>>>(
-                nreq, e->d_req_ids, e->base, e->d_cinfo, e->d_req_info);
+                nreq, e->d_req_ids, e->base, e->d_sigma_inv, e->d_cinfo,
+                e->d_req_info);
             rccl_alltoallv(e, e->d_req_info, req_off.data(), e->d_rc_info,
                            rc_bounds.data(), sizeof(Info16), ncclChar, 1);
             HIP_CHECK(hipMemsetAsync(e->d_rcu, 0,
@@ -1554,7 +1670,8 @@ double mv_engine_run(mv_engine *e, double lower, double thresh,
                     e->d_ghost_comm, e->d_vdeg, e->d_sigma, e->d_cinfo,
                     e->d_cupd, e->d_rc_ids, nrc, e->d_rc_info, e->d_rcu,
                     constant, d_target, e->d_cw, e->d_spill_k, e->d_spill_a,
-                    e->d_spill_off);
+                    e->d_spill_off,
+                    getenv("MV_DBG_SKIP") ? atoi(getenv("MV_DBG_SKIP")) : 0);
         };
         auto dispatch_slots = [&](auto unit_tag) {
             switch (slots) {
@@ -1585,7 +1702,8 @@ double mv_engine_run(mv_engine *e, double lower, double thresh,
                 e->d_rcu, constant, d_target, e->d_cw, e->d_hash_off,
                 e->d_hkeys, e->d_hacc);
         }
-        if (numIters == 1 && e->rows_sorted) {
+        static const bool no_iter1 = getenv("MV_NO_ITER1") != nullptr;
+        if (numIters == 1 && e->rows_sorted && !no_iter1) {
             if (e->unit_weights)
                 launch_iter1(std::integral_constant<bool, true>{});
             else
@@ -1596,10 +1714,27 @@ double mv_engine_run(mv_engine *e, double lower, double thresh,
             dispatch_slots(std::integral_constant<bool, false>{});
         }
         HIP_CHECK(hipEventRecord(ev1, st));
+        if (getenv("MV_MOD_DEBUG")) {
+            HIP_CHECK(hipStreamSynchronize(st));
+            std::vector<Cinfo> cu(lnv);
+            HIP_CHECK(hipMemcpy(cu.data(), e->d_cupd, sizeof(Cinfo) * lnv,
+                                hipMemcpyDeviceToHost));
+            double sd = 0; i64 ssz = 0, nz = 0;
+            for (auto &c : cu) {
+                sd += std::fabs(c.degree);
+                ssz += std::llabs(c.size);
+                nz += (c.size != 0);
+            }
+            std::fprintf(stderr,
+                         "[cupd] iter=%d sum|deg|=%.0f sum|sz|=%lld nz=%lld\n",
+                         numIters, sd, (long long)ssz, (long long)nz);
+        }
+        PHASE("sweep-done");
         sweep_ev.push_back(ev0);
         sweep_ev.push_back(ev1);
         e->stats.sweep_launches++;
 
+        // (phase markers around K6/K7 below)
         // ---- K6 (dspl.hpp:458-471); fused with K7 at p==1 ----
         if (p == 1) {
             k67_apply_and_partials<<<nblocks, 256, 0, st>>>(
@@ -1616,7 +1751,7 @@ double mv_engine_run(mv_engine *e, double lower, double thresh,
                            req_off.data(), sizeof(Info16), ncclChar, 1);
             k_apply_deltas<<<grid_for(std::max<i64>(req_off[p], 1)), 256, 0,
                              st>>>(req_off[p], e->d_req_ids, e->base,
-                                   e->d_req_info, e->d_cinfo);
+                                   e->d_sigma_inv, e->d_req_info, e->d_cinfo);
             HIP_CHECK(hipStreamSynchronize(st));
             e->stats.halo_ms +=
                 std::chrono::duration<double, std::milli>(
@@ -1635,6 +1770,7 @@ double mv_engine_run(mv_engine *e, double lower, double thresh,
             };
             k_partial_sum2<<<nblocks, 256, 0, st>>>(lnv, f, e->d_partials);
         }
+        PHASE("k67-done");
         HIP_CHECK(hipMemcpyAsync(partials.data(), e->d_partials, 16 * nblocks,
                                  hipMemcpyDeviceToHost, st));
         HIP_CHECK(hipStreamSynchronize(st));
@@ -1654,6 +1790,9 @@ double mv_engine_run(mv_engine *e, double lower, double thresh,
             HIP_CHECK(hipStreamSynchronize(st));
         }
         currMod = std::fabs(red[0] * constant - red[1] * constant * constant);
+        if (getenv("MV_MOD_DEBUG"))
+            std::fprintf(stderr, "[mod] iter=%d le=%.17g la=%.17g\n",
+                         numIters, red[0], red[1]);
 
         // ---- trace ----
         if (e->trace_mod && numIters <= e->trace_cap)
