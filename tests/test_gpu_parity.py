@@ -201,3 +201,32 @@ def test_skewed_degree_graph_parity():
     for k in range(min(iters, 300)):
         assert sha(tt[k]) == sha(ott[k]), f"iteration {k+1}"
     print("skewed: iters", iters, "sweep_ms", st["sweep_ms"])
+
+
+def test_binary_file_path_parity():
+    """config-5 style path: write .bin, read back partitioned (-b balanced),
+    run the engine on the read graph — must match the in-memory graph's
+    pinned result (exercises mv_graph_read_binary + a hint-less engine
+    load, i.e. the degree-sorted internal order)."""
+    import json
+    import tempfile
+    from minivite_amd import Graph, Engine
+    from oracle.oracle import sha
+    pins = json.load(open(GOLDEN))
+    pin = pins["rgg_n16384_p1_unit"]
+    g0 = Graph.rgg(16384, 0, 1)
+    with tempfile.TemporaryDirectory() as d:
+        path = os.path.join(d, "g.bin")
+        g0.write_binary(path)
+        g = Graph.read_binary(path, 0, 1)
+        e = Engine(device=0)
+        e.load_graph(g)
+        e.set_trace(64)
+        mod, iters = e.run()
+        tt, _ = e.trace(iters)
+        e.destroy()
+        g.free()
+    g0.free()
+    assert iters == pin["iters"]
+    assert float(mod).hex() == pin["final_mod_hex"]
+    assert [sha(tt[k]) for k in range(iters)] == pin["iter_target_sha"]
