@@ -230,3 +230,46 @@ def test_binary_file_path_parity():
     assert iters == pin["iters"]
     assert float(mod).hex() == pin["final_mod_hex"]
     assert [sha(tt[k]) for k in range(iters)] == pin["iter_target_sha"]
+
+
+def test_large_graph_parity_n262144():
+    """Bigger-than-pin parity: engine vs oracle on the same n=262144 product
+    graph (2.4M directed edges; oracle runs it in a few seconds)."""
+    import numpy as np
+    from minivite_amd import Graph, Engine
+    from oracle.oracle import OracleGraph, louvain, sha
+    nv = 262144
+    g = Graph.rgg(nv, 0, 1)
+    xadj, tails, w = g.arrays()
+    og = OracleGraph.from_csr(nv, 1, np.array([0, nv], dtype=np.int64),
+                              [(xadj, tails, w)])
+    omod, oiters, ott, otm = louvain(og, trace=True)
+    og.free()
+    e = Engine(device=0)
+    e.load_graph(g)
+    e.set_trace(64)
+    mod, iters = e.run()
+    tt, tm = e.trace(iters)
+    e.destroy()
+    g.free()
+    assert iters == oiters
+    assert float(mod).hex() == float(omod).hex()
+    assert [float(m).hex() for m in tm] == [float(m).hex() for m in otm]
+    for k in range(iters):
+        assert sha(tt[k]) == sha(ott[k]), f"iteration {k+1}"
+
+
+def test_cli_binary():
+    """mv355 drop-in CLI reproduces the reference's pinned stdout result."""
+    import re
+    import subprocess
+    repo = os.path.dirname(os.path.dirname(os.path.abspath(__file__)))
+    out = subprocess.run([os.path.join(repo, "minivite_amd", "mv355"),
+                          "-n", "16384", "-l"],
+                         capture_output=True, text=True, timeout=300)
+    assert out.returncode == 0, out.stderr
+    m = re.search(r"Modularity, #Iterations: ([\d.]+), (\d+)", out.stdout)
+    assert m, out.stdout
+    assert m.group(1) == "0.752138"  # main.cpp:193 6-sig-fig format
+    assert m.group(2) == "14"
+    assert "64-bit datatype" in out.stdout
