@@ -273,3 +273,39 @@ def test_cli_binary():
     assert m.group(1) == "0.752138"  # main.cpp:193 6-sig-fig format
     assert m.group(2) == "14"
     assert "64-bit datatype" in out.stdout
+
+
+def test_reference_binary_consumes_framework_bin():
+    """Drop-in cross-check with the REAL reference: write a framework .bin,
+    run the reference binary (oracle/_ref, built from /root/reference
+    sources) on it via -f, and require its printed modularity/iterations to
+    match the engine on the same graph. Skips where the reference binary or
+    MPI runtime is absent."""
+    import re
+    import subprocess
+    import tempfile
+    repo = os.path.dirname(os.path.dirname(os.path.abspath(__file__)))
+    ref = os.path.join(repo, "oracle", "_ref", "minivite")
+    mpiexec = "/opt/conda/bin/mpiexec"
+    if not (os.path.exists(ref) and os.path.exists(mpiexec)):
+        pytest.skip("reference binary or mpiexec not available")
+    from minivite_amd import Graph, Engine
+    g = Graph.rgg(65536, 0, 1)
+    e = Engine(device=0)
+    e.load_graph(g)
+    mod, iters = e.run()
+    env = dict(os.environ)
+    env["LD_PRELOAD"] = "/usr/lib/x86_64-linux-gnu/libstdc++.so.6"
+    with tempfile.TemporaryDirectory() as d:
+        path = os.path.join(d, "g.bin")
+        g.write_binary(path)
+        out = subprocess.run([mpiexec, "-n", "1", ref, "-f", path],
+                             capture_output=True, text=True, timeout=600,
+                             env=env)
+    e.destroy()
+    g.free()
+    assert out.returncode == 0, out.stderr[-500:]
+    m = re.search(r"Modularity, #Iterations: ([\d.e+-]+), (\d+)", out.stdout)
+    assert m, out.stdout
+    assert int(m.group(2)) == iters
+    assert abs(float(m.group(1)) - mod) < 5e-7  # stdout is 6 sig figs
