@@ -27,7 +27,7 @@ def _rank_main(rank, world, nv, out_q):
     import torch
     import torch.distributed as td
     os.environ.setdefault("MASTER_ADDR", "127.0.0.1")
-    os.environ.setdefault("MASTER_PORT", "29601")
+    os.environ.setdefault("MASTER_PORT", str(29601 + world))
     td.init_process_group("gloo", rank=rank, world_size=world)
     sys.path.insert(0, REPO)
     from minivite_amd import Graph
@@ -242,19 +242,19 @@ def _rank_main(rank, world, nv, out_q):
     td.destroy_process_group()
 
 
-@pytest.mark.parametrize("nv", [16384])
-def test_gloo_world2_protocol_matches_pin(nv):
+@pytest.mark.parametrize("nv,world", [(16384, 2), (16384, 4)])
+def test_gloo_protocol_matches_pin(nv, world):
     import torch.multiprocessing as mp
     pins = json.load(open(GOLDEN))
-    pin = pins[f"rgg_n{nv}_p2_unit"]
+    pin = pins[f"rgg_n{nv}_p{world}_unit"]
     ctx = mp.get_context("spawn")
     q = ctx.Queue()
-    procs = [ctx.Process(target=_rank_main, args=(r, 2, nv, q))
-             for r in range(2)]
+    procs = [ctx.Process(target=_rank_main, args=(r, world, nv, q))
+             for r in range(world)]
     for p in procs:
         p.start()
     res = {}
-    for _ in range(2):
+    for _ in range(world):
         r = q.get(timeout=600)
         res[r[0]] = r[1:]
     for p in procs:
