@@ -208,3 +208,9 @@ class Engine:
         if getattr(self, "h", None):
             lib().mv_engine_destroy(self.h)
             self.h = None
+
+    def __del__(self):
+        try:
+            self.destroy()
+        except Exception:
+            pass
