@@ -505,17 +505,13 @@ __global__ __launch_bounds__(256) void k4_sweep(
                 }
 #pragma unroll
                 for (int j = 0; j < CH; j++)
-                    cb[j] = (dbg_skip & 32)
-                                ? (tb[j] << 32) // DEBUG: skip gather
-                                : (tb[j] < lnv)
-                                      ? currComm[tb[j]]
-                                      : ghost_comm[clamp0(tb[j] - lnv)];
+                    cb[j] = (tb[j] < lnv) ? currComm[tb[j]]
+                                          : ghost_comm[clamp0(tb[j] - lnv)];
                 for (int j = 0; j < m; j++) {
                     const i64 tidx = tb[j];
                     const double w = UNIT ? 1.0 : wb[j];
                     if (tidx == i) selfLoop += w; // dspl.hpp:247-248
                     const i64 tcomm = cb[j];
-                    if (dbg_skip & 64) { c0 += w + (double)(tcomm & 1); continue; } // DEBUG: no probes
                     if (tcomm == cc) { c0 += w; continue; }
                     bool found = false;
                     for (int t = 0; t < ns; t++) {
