@@ -144,3 +144,48 @@ def test_rgg_p_reciprocal_edges_consistent():
     finally:
         for g in gs:
             g.free()
+
+
+def test_balanced_read_matches_reference():
+    """-b end-to-end vs the REAL reference: run the reference binary with
+    -f file -b at p=4 and require identical modularity/iterations to the
+    oracle running on OUR balanced-read partitions of the same file.
+    Dev-container only (needs oracle/_ref + conda MPI)."""
+    import re
+    import subprocess
+    repo = os.path.dirname(os.path.dirname(os.path.abspath(__file__)))
+    ref = os.path.join(repo, "oracle", "_ref", "minivite")
+    mpiexec = "/opt/conda/bin/mpiexec"
+    if not (os.path.exists(ref) and os.path.exists(mpiexec)):
+        pytest.skip("reference binary or mpiexec not available")
+    from oracle.oracle import OracleGraph, louvain
+    import ctypes
+    from minivite_amd import lib
+    g = Graph.rgg(16384, 0, 1)
+    with tempfile.TemporaryDirectory() as d:
+        path = os.path.join(d, "g.bin")
+        g.write_binary(path)
+        env = dict(os.environ)
+        env["LD_PRELOAD"] = "/usr/lib/x86_64-linux-gnu/libstdc++.so.6"
+        out = subprocess.run([mpiexec, "-n", "4", ref, "-f", path, "-b"],
+                             capture_output=True, text=True, timeout=600,
+                             env=env)
+        assert out.returncode == 0, out.stderr[-400:]
+        m = re.search(r"Modularity, #Iterations: ([\d.e+-]+), (\d+)",
+                      out.stdout)
+        assert m, out.stdout
+        # our balanced read at p=4 -> oracle on the same partition
+        csrs, parts = [], None
+        for r in range(4):
+            gr = Graph.read_binary(path, r, 4, balanced=True)
+            csrs.append(gr.arrays())
+            if parts is None:
+                pp = lib().mv_graph_parts(gr.h)
+                parts = np.array([pp[k] for k in range(5)], dtype=np.int64)
+            gr.free()
+        og = OracleGraph.from_csr(16384, 4, parts, csrs)
+        mod, iters = louvain(og)
+        og.free()
+    g.free()
+    assert iters == int(m.group(2))
+    assert abs(mod - float(m.group(1))) < 5e-7
