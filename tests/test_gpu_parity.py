@@ -309,3 +309,26 @@ def test_reference_binary_consumes_framework_bin():
     assert m, out.stdout
     assert int(m.group(2)) == iters
     assert abs(float(m.group(1)) - mod) < 5e-7  # stdout is 6 sig figs
+
+
+def test_grid_stride_parity_n4194304():
+    """n=2^22 parity vs oracle — covers the sweep's grid-stride path
+    (lnv > grid*256) and the spill indexing under it; the oracle leg takes
+    ~1 minute of CPU."""
+    import numpy as np
+    from minivite_amd import Graph, Engine
+    from oracle.oracle import OracleGraph, louvain, sha
+    nv = 1 << 22
+    g = Graph.rgg(nv, 0, 1)
+    xadj, tails, w = g.arrays()
+    og = OracleGraph.from_csr(nv, 1, np.array([0, nv], dtype=np.int64),
+                              [(xadj, tails, w)])
+    omod, oiters = louvain(og)
+    e = Engine(device=0)
+    e.load_graph(g)
+    mod, iters = e.run()
+    e.destroy()
+    g.free()
+    og.free()
+    assert iters == oiters
+    assert float(mod).hex() == float(omod).hex()
