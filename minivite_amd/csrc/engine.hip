@@ -1470,54 +1470,6 @@ double mv_engine_run(mv_engine *e, double lower, double thresh,
                                                 e->d_curr, e->d_past);
     HIP_CHECK(hipStreamSynchronize(st));
     PHASE("k3-done");
-    if (getenv("MV_VERIFY")) {
-        HIP_CHECK(hipStreamSynchronize(st));
-        std::vector<i64> hc(lnv);
-        HIP_CHECK(hipMemcpy(hc.data(), e->d_curr, 8 * lnv,
-                            hipMemcpyDeviceToHost));
-        std::vector<unsigned> hs(lnv);
-        HIP_CHECK(hipMemcpy(hs.data(), e->d_sigma, 4 * lnv,
-                            hipMemcpyDeviceToHost));
-        i64 bad = 0;
-        for (i64 k = 0; k < lnv; k++) {
-            const i64 lab = hc[k] >> 32, sl = hc[k] & 0xFFFFFFFFll;
-            if (lab != (i64)hs[k] + e->base || sl != k)
-                if (bad++ < 3)
-                    std::fprintf(stderr, "VERIFY curr[%lld]=%llx exp label "
-                                 "%lld\n", (long long)k,
-                                 (unsigned long long)hc[k],
-                                 (long long)((i64)hs[k] + e->base));
-        }
-        std::fprintf(stderr, "VERIFY curr: %lld bad of %lld\n",
-                     (long long)bad, (long long)lnv);
-        // SELL tidx bounds
-        std::vector<int> st_h(e->sell_elems);
-        HIP_CHECK(hipMemcpy(st_h.data(), e->d_sell_tidx, 4 * e->sell_elems,
-                            hipMemcpyDeviceToHost));
-        std::vector<i64> co(e->nchunks + 1);
-        HIP_CHECK(hipMemcpy(co.data(), e->d_chunk_off, 8 * (e->nchunks + 1),
-                            hipMemcpyDeviceToHost));
-        std::vector<unsigned> dg(lnv), pm(lnv);
-        HIP_CHECK(hipMemcpy(dg.data(), e->d_deg, 4 * lnv,
-                            hipMemcpyDeviceToHost));
-        HIP_CHECK(hipMemcpy(pm.data(), e->d_perm, 4 * lnv,
-                            hipMemcpyDeviceToHost));
-        bad = 0;
-        for (i64 sp = 0; sp < lnv; sp++) {
-            const i64 iint = pm[sp];
-            const i64 eb2 = co[sp >> 6] + (sp & 63);
-            for (unsigned k = 0; k < dg[iint]; k++) {
-                const int ti = st_h[eb2 + (i64)k * 64];
-                if (ti < 0 || ti >= lnv + e->nghost)
-                    if (bad++ < 3)
-                        std::fprintf(stderr, "VERIFY sell s=%lld k=%u ti=%d\n",
-                                     (long long)sp, k, ti);
-            }
-        }
-        std::fprintf(stderr, "VERIFY sell: %lld bad (elems %lld)\n",
-                     (long long)bad, (long long)e->sell_elems);
-        std::fflush(stderr);
-    }
     e->stats.setup_ms =
         std::chrono::duration<double, std::milli>(
             std::chrono::steady_clock::now() - t_setup0)
@@ -1714,21 +1666,6 @@ double mv_engine_run(mv_engine *e, double lower, double thresh,
             dispatch_slots(std::integral_constant<bool, false>{});
         }
         HIP_CHECK(hipEventRecord(ev1, st));
-        if (getenv("MV_MOD_DEBUG")) {
-            HIP_CHECK(hipStreamSynchronize(st));
-            std::vector<Cinfo> cu(lnv);
-            HIP_CHECK(hipMemcpy(cu.data(), e->d_cupd, sizeof(Cinfo) * lnv,
-                                hipMemcpyDeviceToHost));
-            double sd = 0; i64 ssz = 0, nz = 0;
-            for (auto &c : cu) {
-                sd += std::fabs(c.degree);
-                ssz += std::llabs(c.size);
-                nz += (c.size != 0);
-            }
-            std::fprintf(stderr,
-                         "[cupd] iter=%d sum|deg|=%.0f sum|sz|=%lld nz=%lld\n",
-                         numIters, sd, (long long)ssz, (long long)nz);
-        }
         PHASE("sweep-done");
         sweep_ev.push_back(ev0);
         sweep_ev.push_back(ev1);

@@ -98,8 +98,23 @@ int main(int argc, char *argv[]) {
         std::printf("Time to read input file and create distributed graph "
                     "(in s): %f\n",
                     tgen);
-    if (showGraph)
-        std::printf("(-s graph dump not implemented in mv355)\n");
+    if (showGraph) {
+        // graph.hpp:206-248 format: "<global head> <tail> <weight>" per
+        // edge, rank blocks in order
+        for (int r = 0; r < ngpus; r++) {
+            std::printf("###############\nProcess #%d: \n###############\n",
+                        r);
+            const int64_t *xadj = mv_graph_xadj(graphs[r]);
+            const int64_t *tails = mv_graph_tails(graphs[r]);
+            const double *wts = mv_graph_weights(graphs[r]);
+            const int64_t base = mv_graph_parts(graphs[r])[r];
+            const int64_t lnv = mv_graph_lnv(graphs[r]);
+            for (int64_t i = 0; i < lnv; i++)
+                for (int64_t e = xadj[i]; e < xadj[i + 1]; e++)
+                    std::printf("%lld %lld %g\n", (long long)(i + base),
+                                (long long)tails[e], wts[e]);
+        }
+    }
 
     // ---- engines + run (one thread per GPU) ----
     unsigned char cid[MV_COMM_ID_BYTES] = {0};
