@@ -263,6 +263,8 @@ static void evec_push(EVec *e, int64_t row, int64_t tail, double w)
 
 void *orc_rgg_generate(int64_t nv, int p, int unit_weight)
 {
+    /* mirror the reference's input checks (graph.hpp:610-626) */
+    if (p <= 0 || (p & (p - 1)) || nv % p) return NULL;
     OrcGraph *g = orc_graph_new(nv, p);
     const int64_t n_ = nv / p;                       /* graph.hpp:608 */
     const double rc = sqrt(log((double)nv) / (3.14159 * (double)nv)); /* graph.hpp:629, PI utils.hpp:44 */

@@ -35,6 +35,8 @@ CONFIGS = [
     (16384, 2, False),
     (65536, 1, True),
     (65536, 4, True),
+    (32768, 8, True),
+    (16384, 16, True),
 ]
 
 REF_HARNESS = os.path.join(REPO, "oracle", "_ref", "harness_dump")

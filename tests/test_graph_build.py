@@ -19,6 +19,8 @@ from oracle.oracle import OracleGraph
     (16384, 2, True),
     (16384, 4, True),
     (16384, 8, True),
+    (16384, 16, True),
+    (32768, 8, True),
     (16384, 2, False),
 ])
 def test_rgg_bit_identical_to_oracle(nv, p, unit):
