@@ -977,6 +977,7 @@ struct mv_engine {
 };
 
 static void build_sell(mv_engine *e);
+static void free_graph_state(mv_engine *e);
 
 extern "C" {
 
@@ -1022,22 +1023,55 @@ void mv_engine_destroy(mv_engine *e) {
     hipSetDevice(e->device);
     for (auto ev : e->ev_pool) hipEventDestroy(ev);
     if (e->comm) ncclCommDestroy(e->comm);
-    for (void *p : {(void *)e->d_xadj, (void *)e->d_tails, (void *)e->d_ew,
-                    (void *)e->d_sigma, (void *)e->d_sigma_inv,
-                    (void *)e->d_sell_tidx, (void *)e->d_sell_w,
-                    (void *)e->d_deg, (void *)e->d_iota, (void *)e->d_perm,
-                    (void *)e->d_chunk_off, (void *)e->d_curr,
-                    (void *)e->d_past, (void *)e->d_target, (void *)e->d_vdeg,
-                    (void *)e->d_cw, (void *)e->d_cinfo, (void *)e->d_cupd,
-                    (void *)e->d_spill_k, (void *)e->d_spill_a,
-                    (void *)e->d_trace_tmp})
-        if (p) hipFree(p);
+    free_graph_state(e);
     if (e->stream) hipStreamDestroy(e->stream);
     delete e;
 }
 
+// free everything a previous load_graph/run allocated (engines may be
+// re-pointed at a new graph)
+static void free_graph_state(mv_engine *e) {
+    for (void **p : {(void **)&e->d_parts, (void **)&e->d_xadj,
+                     (void **)&e->d_tails, (void **)&e->d_ew,
+                     (void **)&e->d_sigma, (void **)&e->d_sigma_inv,
+                     (void **)&e->d_deg, (void **)&e->d_iota,
+                     (void **)&e->d_perm, (void **)&e->d_chunk_off,
+                     (void **)&e->d_sell_tidx, (void **)&e->d_sell_w,
+                     (void **)&e->d_curr, (void **)&e->d_past,
+                     (void **)&e->d_target, (void **)&e->d_vdeg,
+                     (void **)&e->d_cw, (void **)&e->d_cinfo,
+                     (void **)&e->d_cupd, (void **)&e->d_partials,
+                     (void **)&e->d_red, (void **)&e->d_count,
+                     (void **)&e->d_bounds, (void **)&e->d_ghosts,
+                     (void **)&e->d_ghost_comm, (void **)&e->d_svdata,
+                     (void **)&e->d_svdata_int, (void **)&e->d_scdata,
+                     (void **)&e->d_cand, (void **)&e->d_cand_sorted,
+                     (void **)&e->d_rc_ids, (void **)&e->d_rc_info,
+                     (void **)&e->d_rcu, (void **)&e->d_req_ids,
+                     (void **)&e->d_req_info, (void **)&e->d_cub_tmp,
+                     (void **)&e->d_spill_k, (void **)&e->d_spill_a,
+                     (void **)&e->d_spill_off, (void **)&e->d_hash_off,
+                     (void **)&e->d_hkeys, (void **)&e->d_hacc,
+                     (void **)&e->d_trace_tmp}) {
+        if (*p) {
+            HIP_CHECK(hipFree(*p));
+            *p = nullptr;
+        }
+    }
+    e->sell_elems = 0;
+    e->spill_elems = 0;
+    e->rc_cap = 0;
+    e->req_cap = 0;
+    e->hash_total = 0;
+    e->nghost = 0;
+    e->ssz = 0;
+    e->nhi = 0;
+    e->cub_tmp_bytes = 0;
+}
+
 int mv_engine_load_graph(mv_engine *e, const mv_graph *g) {
     HIP_CHECK(hipSetDevice(e->device));
+    if (e->d_xadj) free_graph_state(e); // re-load on a live engine
     e->nv = mv_graph_nv(g);
     if (e->nv >= (1ll << 31)) { // community-handle packing needs 31-bit ids
         std::fprintf(stderr, "mv_engine_load_graph: nv >= 2^31 unsupported\n");

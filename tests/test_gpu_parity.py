@@ -388,3 +388,24 @@ def test_adversarial_graphs_parity():
             assert abs(mod - omod) < 1e-9, f"trial {trial}"
         for k in range(min(iters, 256)):
             assert sha(tt[k]) == sha(ott[k]), f"trial {trial} iter {k+1}"
+
+
+def test_engine_graph_reload():
+    """load_graph on a live engine replaces the previous graph cleanly."""
+    import json
+    from minivite_amd import Graph, Engine
+    pins = json.load(open(GOLDEN))
+    e = Engine(device=0)
+    g1 = Graph.rgg(65536, 0, 1)
+    e.load_graph(g1)
+    m1, i1 = e.run()
+    g2 = Graph.rgg(16384, 0, 1)
+    e.load_graph(g2)
+    m2, i2 = e.run()
+    e.destroy()
+    g1.free()
+    g2.free()
+    assert float(m1).hex() == pins["rgg_n65536_p1_unit"]["final_mod_hex"]
+    assert float(m2).hex() == pins["rgg_n16384_p1_unit"]["final_mod_hex"]
+    assert (i1, i2) == (pins["rgg_n65536_p1_unit"]["iters"],
+                        pins["rgg_n16384_p1_unit"]["iters"])
