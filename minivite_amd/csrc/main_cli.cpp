@@ -21,6 +21,7 @@
 #include <cstring>
 #include <string>
 #include <thread>
+#include <cmath>
 #include <unistd.h>
 #include <vector>
 
@@ -98,6 +99,34 @@ int main(int argc, char *argv[]) {
         std::printf("Time to read input file and create distributed graph "
                     "(in s): %f\n",
                     tgen);
+    if (getenv("MV_DIST_STATS")) {
+        // print_dist_stats equivalent (graph.hpp:251-286; the reference
+        // gates it behind -DPRINT_DIST_STATS)
+        long long sumdeg = 0, maxdeg = 0, ne = 0;
+        double sum_sq = 0;
+        for (int r = 0; r < ngpus; r++) {
+            const long long lne = (long long)mv_graph_lne(graphs[r]);
+            sumdeg += lne;
+            if (lne > maxdeg) maxdeg = lne;
+            sum_sq += (double)lne * (double)lne;
+            ne += lne;
+        }
+        const double average = (double)sumdeg / ngpus;
+        const double avg_sq = sum_sq / ngpus;
+        const double var = avg_sq - average * average;
+        std::printf("\n-------------------------------------------------------\n");
+        std::printf("Graph edge distribution characteristics\n");
+        std::printf("-------------------------------------------------------\n");
+        std::printf("Number of vertices: %lld\n",
+                    (long long)mv_graph_nv(graphs[0]));
+        std::printf("Number of edges: %lld\n", ne);
+        std::printf("Maximum number of edges: %lld\n", maxdeg);
+        std::printf("Average number of edges: %g\n", average);
+        std::printf("Expected value of X^2: %g\n", avg_sq);
+        std::printf("Variance: %g\n", var);
+        std::printf("Standard deviation: %g\n", sqrt(var));
+        std::printf("-------------------------------------------------------\n");
+    }
     if (showGraph) {
         // graph.hpp:206-248 format: "<global head> <tail> <weight>" per
         // edge, rank blocks in order
