@@ -29,8 +29,9 @@
 //    lookup (dspl.hpp:253-260). Unit-weight graphs skip the weight stream
 //    entirely (every w == 1.0, detected at load).
 //  * Community info (localCinfo/localCupdate, dspl.hpp:61-66) is AoS
-//    {int64 size; double degree} indexed by community label - base, so one
-//    lookup costs one cache line, not two.
+//    {int64 size; double degree} in INTERNAL order, addressed through
+//    packed community handles (see h_label/h_slot below), so one lookup
+//    costs one spatially-local cache line.
 //
 // FP discipline: built with -ffp-contract=off so the dQ gain expression
 // (dspl.hpp:212) and all accumulations carry the same bits as the

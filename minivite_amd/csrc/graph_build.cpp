@@ -411,9 +411,6 @@ mv_graph *mv_graph_read_binary(const char *path, int rank, int nranks,
     const int64_t v0 = g->parts[rank], v1 = g->parts[rank + 1];
     const int64_t lnv = v1 - v0;
     g->xadj.resize(lnv + 1);
-#ifdef _WIN32
-#error unsupported
-#endif
     // offsets slice
     std::fseek(f, (long)(16 + v0 * 8), SEEK_SET);
     if (std::fread(g->xadj.data(), 8, lnv + 1, f) != (size_t)(lnv + 1)) {

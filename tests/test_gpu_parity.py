@@ -260,12 +260,14 @@ def test_large_graph_parity_n262144():
 
 
 def test_cli_binary():
-    """mv355 drop-in CLI reproduces the reference's pinned stdout result."""
+    """mv355 drop-in CLI reproduces the reference's pinned stdout result,
+    on both the generator (-n -l) and the file (-f) input paths."""
     import re
     import subprocess
+    import tempfile
     repo = os.path.dirname(os.path.dirname(os.path.abspath(__file__)))
-    out = subprocess.run([os.path.join(repo, "minivite_amd", "mv355"),
-                          "-n", "16384", "-l"],
+    cli = os.path.join(repo, "minivite_amd", "mv355")
+    out = subprocess.run([cli, "-n", "16384", "-l"],
                          capture_output=True, text=True, timeout=300)
     assert out.returncode == 0, out.stderr
     m = re.search(r"Modularity, #Iterations: ([\d.]+), (\d+)", out.stdout)
@@ -273,6 +275,18 @@ def test_cli_binary():
     assert m.group(1) == "0.752138"  # main.cpp:193 6-sig-fig format
     assert m.group(2) == "14"
     assert "64-bit datatype" in out.stdout
+    # -f path on a framework-written bin
+    from minivite_amd import Graph
+    g = Graph.rgg(16384, 0, 1)
+    with tempfile.TemporaryDirectory() as d:
+        path = os.path.join(d, "g.bin")
+        g.write_binary(path)
+        out2 = subprocess.run([cli, "-f", path], capture_output=True,
+                              text=True, timeout=300)
+    g.free()
+    assert out2.returncode == 0, out2.stderr
+    m2 = re.search(r"Modularity, #Iterations: ([\d.]+), (\d+)", out2.stdout)
+    assert m2 and m2.group(1) == "0.752138" and m2.group(2) == "14", out2.stdout
 
 
 def test_reference_binary_consumes_framework_bin():
