@@ -1021,11 +1021,11 @@ mv_engine *mv_engine_create(int device, int rank, int nranks,
 
 void mv_engine_destroy(mv_engine *e) {
     if (!e) return;
-    hipSetDevice(e->device);
-    for (auto ev : e->ev_pool) hipEventDestroy(ev);
+    (void)hipSetDevice(e->device); // teardown: best effort
+    for (auto ev : e->ev_pool) (void)hipEventDestroy(ev);
     if (e->comm) ncclCommDestroy(e->comm);
     free_graph_state(e);
-    if (e->stream) hipStreamDestroy(e->stream);
+    if (e->stream) (void)hipStreamDestroy(e->stream);
     delete e;
 }
 
@@ -1237,12 +1237,12 @@ static void build_sell(mv_engine *e) {
         unsigned *d_degs = nullptr;
         HIP_CHECK(hipMalloc(&d_degs, 4 * std::max<i64>(lnv, 1)));
         size_t tb = 0;
-        hipcub::DeviceRadixSort::SortPairsDescending(
+        (void)hipcub::DeviceRadixSort::SortPairsDescending(
             nullptr, tb, e->d_deg, d_degs, e->d_iota, e->d_perm, lnv, 0,
             32, st);
         void *d_tmp = nullptr;
         HIP_CHECK(hipMalloc(&d_tmp, std::max<size_t>(tb, 1)));
-        hipcub::DeviceRadixSort::SortPairsDescending(
+        (void)hipcub::DeviceRadixSort::SortPairsDescending(
             d_tmp, tb, e->d_deg, d_degs, e->d_iota, e->d_perm, lnv, 0, 32,
             st);
         // high-degree split + per-vertex hash regions (unit weights
@@ -1289,12 +1289,12 @@ static void build_sell(mv_engine *e) {
             e->nchunks, lnv, e->d_perm, e->d_deg, d_sizes);
         HIP_CHECK(hipMemsetAsync(e->d_chunk_off, 0, 8, st));
         size_t tb2 = 0;
-        hipcub::DeviceScan::InclusiveSum(nullptr, tb2, d_sizes,
+        (void)hipcub::DeviceScan::InclusiveSum(nullptr, tb2, d_sizes,
                                          e->d_chunk_off + 1, e->nchunks,
                                          st);
         void *d_tmp2 = nullptr;
         HIP_CHECK(hipMalloc(&d_tmp2, std::max<size_t>(tb2, 1)));
-        hipcub::DeviceScan::InclusiveSum(d_tmp2, tb2, d_sizes,
+        (void)hipcub::DeviceScan::InclusiveSum(d_tmp2, tb2, d_sizes,
                                          e->d_chunk_off + 1, e->nchunks,
                                          st);
         i64 total = 0;
@@ -1406,18 +1406,18 @@ double mv_engine_run(mv_engine *e, double lower, double thresh,
                 HIP_CHECK(hipMalloc(&e->d_ghosts,
                                     8 * std::max<i64>((i64)nrem, 1)));
             size_t tmp1 = 0, tmp2 = 0;
-            hipcub::DeviceRadixSort::SortKeys(nullptr, tmp1, d_rem, d_sorted,
+            (void)hipcub::DeviceRadixSort::SortKeys(nullptr, tmp1, d_rem, d_sorted,
                                               (int64_t)nrem, 0, 64, st);
             i64 *d_ng = nullptr;
             HIP_CHECK(hipMalloc(&d_ng, 8));
-            hipcub::DeviceSelect::Unique(nullptr, tmp2, d_sorted, e->d_ghosts,
+            (void)hipcub::DeviceSelect::Unique(nullptr, tmp2, d_sorted, e->d_ghosts,
                                          d_ng, (int64_t)nrem, st);
             size_t tmpb = std::max(tmp1, tmp2);
             void *d_tmp = nullptr;
             HIP_CHECK(hipMalloc(&d_tmp, std::max<size_t>(tmpb, 1)));
-            hipcub::DeviceRadixSort::SortKeys(d_tmp, tmp1, d_rem, d_sorted,
+            (void)hipcub::DeviceRadixSort::SortKeys(d_tmp, tmp1, d_rem, d_sorted,
                                               (int64_t)nrem, 0, 64, st);
-            hipcub::DeviceSelect::Unique(d_tmp, tmp2, d_sorted, e->d_ghosts,
+            (void)hipcub::DeviceSelect::Unique(d_tmp, tmp2, d_sorted, e->d_ghosts,
                                          d_ng, (int64_t)nrem, st);
             HIP_CHECK(hipMemcpyAsync(&e->nghost, d_ng, 8,
                                      hipMemcpyDeviceToHost, st));
@@ -1542,11 +1542,11 @@ double mv_engine_run(mv_engine *e, double lower, double thresh,
                 HIP_CHECK(hipMalloc(&e->d_rc_info, sizeof(Info16) * cand_max));
                 HIP_CHECK(hipMalloc(&e->d_rcu, sizeof(Info16) * cand_max));
                 size_t t1 = 0, t2 = 0;
-                hipcub::DeviceRadixSort::SortKeys(nullptr, t1, e->d_cand,
+                (void)hipcub::DeviceRadixSort::SortKeys(nullptr, t1, e->d_cand,
                                                   e->d_cand_sorted, cand_max, 0,
                                                   64, st);
                 i64 *dummy = nullptr;
-                hipcub::DeviceSelect::Unique(nullptr, t2, e->d_cand_sorted,
+                (void)hipcub::DeviceSelect::Unique(nullptr, t2, e->d_cand_sorted,
                                              e->d_rc_ids, dummy, cand_max, st);
                 size_t need = std::max(t1, t2);
                 if (need > e->cub_tmp_bytes) {
@@ -1567,12 +1567,12 @@ double mv_engine_run(mv_engine *e, double lower, double thresh,
                                      hipMemcpyDeviceToHost, st));
             HIP_CHECK(hipStreamSynchronize(st));
             size_t tb = e->cub_tmp_bytes;
-            hipcub::DeviceRadixSort::SortKeys(e->d_cub_tmp, tb, e->d_cand,
+            (void)hipcub::DeviceRadixSort::SortKeys(e->d_cub_tmp, tb, e->d_cand,
                                               e->d_cand_sorted, (int64_t)ncand,
                                               0, 64, st);
             i64 *d_nrc = (i64 *)e->d_count; // reuse as output slot
             tb = e->cub_tmp_bytes;
-            hipcub::DeviceSelect::Unique(e->d_cub_tmp, tb, e->d_cand_sorted,
+            (void)hipcub::DeviceSelect::Unique(e->d_cub_tmp, tb, e->d_cand_sorted,
                                          e->d_rc_ids, d_nrc, (int64_t)ncand, st);
             HIP_CHECK(hipMemcpyAsync(&nrc, d_nrc, 8, hipMemcpyDeviceToHost, st));
             // ghost communities: labels -> handles for the sweep

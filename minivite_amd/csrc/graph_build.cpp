@@ -189,7 +189,6 @@ mv_graph *mv_graph_rgg(int64_t nv, int rank, int nranks,
     // Tails per row sorted ascending == the reference's (row, tail) sort
     // (graph.hpp:1145-1153) restricted to the row.
     const int64_t own_off = (int64_t)(rank - s_lo) * n_;
-    const int64_t base = g->parts[rank];
 
     // two passes (count, then fill + per-row insertion sort) — no
     // intermediate per-row vectors, so n=2^26 fits comfortably
