@@ -1,0 +1,329 @@
+// sweep_variants.hip — standalone microbenchmark of candidate K4 sweep
+// structures for round 2, on synthetic steady-state-shaped data (NOT wired
+// into the product; results recorded in experiments/RESULTS.md).
+//
+// Synthetic workload mimicking Louvain steady state on an RGG at n=2^24:
+// deg ~ Poisson(10.4) via a fixed per-vertex degree table, neighbor
+// internal indices within a +-8192 window (the sigma spatial layout),
+// community values with ~3 distinct candidates per vertex.
+//
+// Variants:
+//   V0  lane-per-vertex, 8-chunked loads+gathers, LDS linear slots
+//       (replica of the production kernel's skeleton)
+//   V1  two vertices per lane, phases interleaved (ILP experiment)
+//   V3  split: phase A writes tcomm[] SELL-aligned (pure gather kernel),
+//       phase B consumes it coalesced (no gathers in the probe kernel)
+//
+// Build: hipcc --offload-arch=gfx950 -O3 -std=c++17 sweep_variants.hip -o sweep_variants
+// Run:   ./sweep_variants [nv]
+
+#include <chrono>
+#include <cstdint>
+#include <cstdio>
+#include <cstdlib>
+#include <vector>
+
+#include <hip/hip_runtime.h>
+
+#define HC(x)                                                                 \
+    do {                                                                      \
+        hipError_t e_ = (x);                                                  \
+        if (e_ != hipSuccess) {                                               \
+            std::fprintf(stderr, "HIP %s @%d\n", hipGetErrorString(e_),       \
+                         __LINE__);                                           \
+            std::exit(1);                                                     \
+        }                                                                     \
+    } while (0)
+
+using i64 = int64_t;
+constexpr int CH = 8;
+constexpr int SLOTS = 8;
+
+// ---------------- V0: production skeleton ----------------
+__global__ __launch_bounds__(256) void v0(
+    i64 lnv, const unsigned *__restrict__ deg, const i64 *__restrict__ coff,
+    const int *__restrict__ sell, const i64 *__restrict__ comm,
+    const double *__restrict__ aux, i64 *__restrict__ out) {
+    extern __shared__ char smem[];
+    i64 *skey = (i64 *)smem;
+    double *sacc = (double *)(smem + 8 * SLOTS * blockDim.x);
+    const int tid = threadIdx.x;
+    const i64 g0 = blockIdx.x * (i64)blockDim.x + threadIdx.x;
+    const i64 stride = (i64)gridDim.x * blockDim.x;
+    for (i64 s = g0; s < lnv; s += stride) {
+        const int d = (int)deg[s];
+        const i64 eb = coff[s >> 6] + (s & 63);
+        const i64 cc = comm[s];
+        double c0 = 0.0;
+        int ns = 0;
+        for (int k0 = 0; k0 < d; k0 += CH) {
+            const int m = min(CH, d - k0);
+            i64 tb[CH], cb[CH];
+#pragma unroll
+            for (int j = 0; j < CH; j++)
+                tb[j] = sell[(j < m) ? eb + (i64)(k0 + j) * 64 : eb];
+#pragma unroll
+            for (int j = 0; j < CH; j++) cb[j] = comm[tb[j]];
+            for (int j = 0; j < m; j++) {
+                const i64 tc = cb[j];
+                if (tc == cc) { c0 += 1.0; continue; }
+                bool f = false;
+                for (int t = 0; t < ns; t++)
+                    if (skey[t * blockDim.x + tid] == tc) {
+                        sacc[t * blockDim.x + tid] += 1.0;
+                        f = true;
+                        break;
+                    }
+                if (!f && ns < SLOTS) {
+                    skey[ns * blockDim.x + tid] = tc;
+                    sacc[ns * blockDim.x + tid] = 1.0;
+                    ns++;
+                }
+            }
+        }
+        double bg = 0.0;
+        i64 bl = cc;
+        for (int t = 0; t < ns; t++) {
+            const i64 y = skey[t * blockDim.x + tid];
+            const double eiy = sacc[t * blockDim.x + tid];
+            const double g = 2.0 * eiy - 1e-7 * aux[y & (lnv - 1)];
+            if (g > bg) { bg = g; bl = y; }
+        }
+        out[s] = bl + (i64)c0;
+    }
+}
+
+// ---------------- V1: two vertices per lane ----------------
+__global__ __launch_bounds__(256) void v1(
+    i64 lnv, const unsigned *__restrict__ deg, const i64 *__restrict__ coff,
+    const int *__restrict__ sell, const i64 *__restrict__ comm,
+    const double *__restrict__ aux, i64 *__restrict__ out) {
+    extern __shared__ char smem[];
+    i64 *skey = (i64 *)smem; // 2 * SLOTS per lane
+    double *sacc = (double *)(smem + 8 * 2 * SLOTS * blockDim.x);
+    const int tid = threadIdx.x;
+    const i64 half = (lnv + 1) / 2;
+    const i64 g0 = blockIdx.x * (i64)blockDim.x + threadIdx.x;
+    const i64 stride = (i64)gridDim.x * blockDim.x;
+    for (i64 s = g0; s < half; s += stride) {
+        const i64 sA = s, sB = s + half;
+        const bool hasB = sB < lnv;
+        const int dA = (int)deg[sA];
+        const int dB = hasB ? (int)deg[sB] : 0;
+        const i64 ebA = coff[sA >> 6] + (sA & 63);
+        const i64 ebB = hasB ? coff[sB >> 6] + (sB & 63) : 0;
+        const i64 ccA = comm[sA];
+        const i64 ccB = hasB ? comm[sB] : 0;
+        double c0A = 0.0, c0B = 0.0;
+        int nsA = 0, nsB = 0;
+        const int kmax = max(dA, dB);
+        for (int k0 = 0; k0 < kmax; k0 += CH) {
+            const int mA = min(CH, max(dA - k0, 0));
+            const int mB = min(CH, max(dB - k0, 0));
+            i64 tbA[CH], tbB[CH], cbA[CH], cbB[CH];
+#pragma unroll
+            for (int j = 0; j < CH; j++) {
+                tbA[j] = sell[(j < mA) ? ebA + (i64)(k0 + j) * 64 : ebA];
+                tbB[j] = sell[(j < mB) ? ebB + (i64)(k0 + j) * 64 : ebB];
+            }
+#pragma unroll
+            for (int j = 0; j < CH; j++) {
+                cbA[j] = comm[tbA[j]];
+                cbB[j] = comm[tbB[j]];
+            }
+            for (int j = 0; j < mA; j++) {
+                const i64 tc = cbA[j];
+                if (tc == ccA) { c0A += 1.0; continue; }
+                bool f = false;
+                for (int t = 0; t < nsA; t++)
+                    if (skey[t * blockDim.x + tid] == tc) {
+                        sacc[t * blockDim.x + tid] += 1.0;
+                        f = true;
+                        break;
+                    }
+                if (!f && nsA < SLOTS) {
+                    skey[nsA * blockDim.x + tid] = tc;
+                    sacc[nsA * blockDim.x + tid] = 1.0;
+                    nsA++;
+                }
+            }
+            for (int j = 0; j < mB; j++) {
+                const i64 tc = cbB[j];
+                if (tc == ccB) { c0B += 1.0; continue; }
+                bool f = false;
+                for (int t = 0; t < nsB; t++)
+                    if (skey[(SLOTS + t) * blockDim.x + tid] == tc) {
+                        sacc[(SLOTS + t) * blockDim.x + tid] += 1.0;
+                        f = true;
+                        break;
+                    }
+                if (!f && nsB < SLOTS) {
+                    skey[(SLOTS + nsB) * blockDim.x + tid] = tc;
+                    sacc[(SLOTS + nsB) * blockDim.x + tid] = 1.0;
+                    nsB++;
+                }
+            }
+        }
+        double bgA = 0.0, bgB = 0.0;
+        i64 blA = ccA, blB = ccB;
+        for (int t = 0; t < nsA; t++) {
+            const i64 y = skey[t * blockDim.x + tid];
+            const double g = 2.0 * sacc[t * blockDim.x + tid] -
+                             1e-7 * aux[y & (lnv - 1)];
+            if (g > bgA) { bgA = g; blA = y; }
+        }
+        for (int t = 0; t < nsB; t++) {
+            const i64 y = skey[(SLOTS + t) * blockDim.x + tid];
+            const double g = 2.0 * sacc[(SLOTS + t) * blockDim.x + tid] -
+                             1e-7 * aux[y & (lnv - 1)];
+            if (g > bgB) { bgB = g; blB = y; }
+        }
+        out[sA] = blA + (i64)c0A;
+        if (hasB) out[sB] = blB + (i64)c0B;
+    }
+}
+
+// ---------------- V3: split gather / probe ----------------
+__global__ void v3_gather(i64 elems, const int *__restrict__ sell,
+                          const i64 *__restrict__ comm,
+                          i64 *__restrict__ tcomm) {
+    for (i64 x = blockIdx.x * (i64)blockDim.x + threadIdx.x; x < elems;
+         x += (i64)gridDim.x * blockDim.x)
+        tcomm[x] = comm[sell[x]];
+}
+
+__global__ __launch_bounds__(256) void v3_probe(
+    i64 lnv, const unsigned *__restrict__ deg, const i64 *__restrict__ coff,
+    const i64 *__restrict__ tcomm, const i64 *__restrict__ comm,
+    const double *__restrict__ aux, i64 *__restrict__ out) {
+    extern __shared__ char smem[];
+    i64 *skey = (i64 *)smem;
+    double *sacc = (double *)(smem + 8 * SLOTS * blockDim.x);
+    const int tid = threadIdx.x;
+    const i64 g0 = blockIdx.x * (i64)blockDim.x + threadIdx.x;
+    const i64 stride = (i64)gridDim.x * blockDim.x;
+    for (i64 s = g0; s < lnv; s += stride) {
+        const int d = (int)deg[s];
+        const i64 eb = coff[s >> 6] + (s & 63);
+        const i64 cc = comm[s];
+        double c0 = 0.0;
+        int ns = 0;
+        for (int k = 0; k < d; k++) {
+            const i64 tc = tcomm[eb + (i64)k * 64];
+            if (tc == cc) { c0 += 1.0; continue; }
+            bool f = false;
+            for (int t = 0; t < ns; t++)
+                if (skey[t * blockDim.x + tid] == tc) {
+                    sacc[t * blockDim.x + tid] += 1.0;
+                    f = true;
+                    break;
+                }
+            if (!f && ns < SLOTS) {
+                skey[ns * blockDim.x + tid] = tc;
+                sacc[ns * blockDim.x + tid] = 1.0;
+                ns++;
+            }
+        }
+        double bg = 0.0;
+        i64 bl = cc;
+        for (int t = 0; t < ns; t++) {
+            const i64 y = skey[t * blockDim.x + tid];
+            const double g = 2.0 * sacc[t * blockDim.x + tid] -
+                             1e-7 * aux[y & (lnv - 1)];
+            if (g > bg) { bg = g; bl = y; }
+        }
+        out[s] = bl + (i64)c0;
+    }
+}
+
+int main(int argc, char **argv) {
+    const i64 lnv = argc > 1 ? atoll(argv[1]) : (1ll << 24);
+    const i64 nchunks = (lnv + 63) / 64;
+    std::vector<unsigned> deg(lnv);
+    std::vector<i64> coff(nchunks + 1, 0);
+    // deterministic pseudo-Poisson degrees around 10.4, window neighbors
+    uint64_t rng = 0x9E3779B97F4A7C15ull;
+    auto rnd = [&]() { rng ^= rng << 13; rng ^= rng >> 7; rng ^= rng << 17;
+                       return rng; };
+    for (i64 i = 0; i < lnv; i++) deg[i] = 6 + (unsigned)(rnd() % 9);
+    for (i64 c = 0; c < nchunks; c++) {
+        unsigned w = 0;
+        for (i64 s = c * 64; s < std::min(c * 64 + 64, lnv); s++)
+            w = std::max(w, deg[s]);
+        coff[c + 1] = coff[c] + (i64)w * 64;
+    }
+    const i64 elems = coff[nchunks];
+    std::vector<int> sell(elems, 0);
+    std::vector<i64> comm(lnv);
+    for (i64 i = 0; i < lnv; i++) {
+        // ~3 distinct communities in each neighborhood: comm = i rounded to
+        // a window-sized bucket + jitter
+        comm[i] = ((i >> 10) << 10) + (i64)(rnd() % 3) * 341;
+        if (comm[i] >= lnv) comm[i] = lnv - 1;
+    }
+    for (i64 s = 0; s < lnv; s++) {
+        const i64 eb = coff[s >> 6] + (s & 63);
+        for (unsigned k = 0; k < deg[s]; k++) {
+            i64 nb = s + (i64)(rnd() % 16384) - 8192; // spatial window
+            if (nb < 0) nb += lnv;
+            if (nb >= lnv) nb -= lnv;
+            sell[eb + (i64)k * 64] = (int)nb;
+        }
+    }
+    std::vector<double> aux(lnv, 1.0);
+
+    unsigned *d_deg;
+    i64 *d_coff, *d_comm, *d_out, *d_tcomm;
+    int *d_sell;
+    double *d_aux;
+    HC(hipMalloc(&d_deg, 4 * lnv));
+    HC(hipMalloc(&d_coff, 8 * (nchunks + 1)));
+    HC(hipMalloc(&d_sell, 4 * elems));
+    HC(hipMalloc(&d_comm, 8 * lnv));
+    HC(hipMalloc(&d_out, 8 * lnv));
+    HC(hipMalloc(&d_aux, 8 * lnv));
+    HC(hipMalloc(&d_tcomm, 8 * elems));
+    HC(hipMemcpy(d_deg, deg.data(), 4 * lnv, hipMemcpyHostToDevice));
+    HC(hipMemcpy(d_coff, coff.data(), 8 * (nchunks + 1),
+                 hipMemcpyHostToDevice));
+    HC(hipMemcpy(d_sell, sell.data(), 4 * elems, hipMemcpyHostToDevice));
+    HC(hipMemcpy(d_comm, comm.data(), 8 * lnv, hipMemcpyHostToDevice));
+    HC(hipMemcpy(d_aux, aux.data(), 8 * lnv, hipMemcpyHostToDevice));
+
+    const int grid = (int)std::min<i64>((lnv + 255) / 256, 2048);
+    const i64 edges = [&] {
+        i64 t = 0;
+        for (i64 i = 0; i < lnv; i++) t += deg[i];
+        return t;
+    }();
+
+    auto bench = [&](const char *name, auto &&fn) {
+        fn(); // warm
+        HC(hipDeviceSynchronize());
+        const int reps = 20;
+        auto t0 = std::chrono::steady_clock::now();
+        for (int r = 0; r < reps; r++) fn();
+        HC(hipDeviceSynchronize());
+        double ms = std::chrono::duration<double, std::milli>(
+                        std::chrono::steady_clock::now() - t0)
+                        .count() /
+                    reps;
+        std::printf("%-10s %8.3f ms   %7.1f G edge/s\n", name, ms,
+                    edges / ms / 1e6);
+    };
+
+    bench("V0", [&] {
+        v0<<<grid, 256, SLOTS * 256 * 16>>>(lnv, d_deg, d_coff, d_sell,
+                                            d_comm, d_aux, d_out);
+    });
+    bench("V1-2vpl", [&] {
+        v1<<<grid, 256, 2 * SLOTS * 256 * 16>>>(lnv, d_deg, d_coff, d_sell,
+                                                d_comm, d_aux, d_out);
+    });
+    bench("V3-split", [&] {
+        v3_gather<<<2048, 256>>>(elems, d_sell, d_comm, d_tcomm);
+        v3_probe<<<grid, 256, SLOTS * 256 * 16>>>(lnv, d_deg, d_coff, d_tcomm,
+                                                  d_comm, d_aux, d_out);
+    });
+    return 0;
+}
