@@ -915,6 +915,8 @@ struct mv_engine {
     double *d_vdeg = nullptr, *d_cw = nullptr;
     Cinfo *d_cinfo = nullptr, *d_cupd = nullptr;
     double *d_partials = nullptr; // 2 * nblocks
+    double *h_partials = nullptr; // pinned mirror (pageable D2H costs
+                                  // ~40 us/iteration in staging latency)
     double *d_red = nullptr;      // 2 doubles for allreduce
 
     // ghosts / halo
@@ -1059,6 +1061,10 @@ static void free_graph_state(mv_engine *e) {
             *p = nullptr;
         }
     }
+    if (e->h_partials) {
+        HIP_CHECK(hipHostFree(e->h_partials));
+        e->h_partials = nullptr;
+    }
     e->sell_elems = 0;
     e->spill_elems = 0;
     e->rc_cap = 0;
@@ -1148,6 +1154,7 @@ int mv_engine_load_graph(mv_engine *e, const mv_graph *g) {
     HIP_CHECK(hipMalloc(&e->d_red, 16));
     const int nblocks = grid_for(lnv);
     HIP_CHECK(hipMalloc(&e->d_partials, 16 * nblocks));
+    HIP_CHECK(hipHostMalloc(&e->h_partials, 16 * nblocks));
 
     // K4 geometry: 256-thread blocks, 8 LDS slots/lane (32 KiB/block ->
     // 4-5 blocks/CU). Spill sizing: uniform per-thread max_degree extents
@@ -1485,8 +1492,8 @@ double mv_engine_run(mv_engine *e, double lower, double thresh,
         };
         k_partial_sum2<<<nblocks, 256, 0, st>>>(lnv, f, e->d_partials);
     }
-    std::vector<double> partials(2 * nblocks);
-    HIP_CHECK(hipMemcpyAsync(partials.data(), e->d_partials, 16 * nblocks,
+    double *const partials = e->h_partials;
+    HIP_CHECK(hipMemcpyAsync(partials, e->d_partials, 16 * nblocks,
                              hipMemcpyDeviceToHost, st));
     HIP_CHECK(hipStreamSynchronize(st));
     double localW = 0.0;
@@ -1743,7 +1750,7 @@ double mv_engine_run(mv_engine *e, double lower, double thresh,
             k_partial_sum2<<<nblocks, 256, 0, st>>>(lnv, f, e->d_partials);
         }
         PHASE("k67-done");
-        HIP_CHECK(hipMemcpyAsync(partials.data(), e->d_partials, 16 * nblocks,
+        HIP_CHECK(hipMemcpyAsync(partials, e->d_partials, 16 * nblocks,
                                  hipMemcpyDeviceToHost, st));
         HIP_CHECK(hipStreamSynchronize(st));
         double le = 0.0, la = 0.0;
