@@ -873,6 +873,352 @@ __global__ __launch_bounds__(256) void k4_sweep_iter1(
     }
 }
 
+
+// ---- Single-rank (p==1) specializations: u32 LABEL communities ----
+// At one rank every community is local (base 0, no ghosts, no remote
+// paths), so a community value can be its 32-bit LABEL instead of the
+// 64-bit handle: label comparisons keep the reference tie-break and
+// singleton-guard semantics bit-for-bit, and the per-edge community
+// gather — the sweep's dominant random traffic — halves to 4 B. Candidate
+// info stays in internal (spatial) order; the label->slot hop goes
+// through sigma_inv once per DISTINCT candidate, not per edge. Same
+// reference anchors as the generic kernels (dspl.hpp:174-405).
+__global__ void k3_init_comm32(i64 lnv, const unsigned *__restrict__ sigma,
+                               unsigned *__restrict__ curr,
+                               unsigned *__restrict__ past) {
+    for (i64 k = blockIdx.x * (i64)blockDim.x + threadIdx.x; k < lnv;
+         k += (i64)gridDim.x * blockDim.x) {
+        const unsigned l = sigma[k]; // label of the vertex slot k holds
+        curr[k] = l;
+        past[k] = l;
+    }
+}
+
+__global__ void k_depermute32(i64 lnv,
+                              const unsigned *__restrict__ sigma_inv,
+                              const unsigned *__restrict__ in,
+                              i64 *__restrict__ out) {
+    for (i64 v = blockIdx.x * (i64)blockDim.x + threadIdx.x; v < lnv;
+         v += (i64)gridDim.x * blockDim.x)
+        out[v] = (i64)in[sigma_inv[v]];
+}
+
+template <int SLOTS, bool UNIT>
+__global__ __launch_bounds__(256) void k4_sweep_p1(
+    i64 s_begin, i64 lnv, const unsigned *__restrict__ perm,
+    const unsigned *__restrict__ deg_int, const i64 *__restrict__ chunk_off,
+    const int *__restrict__ sell_tidx, const double *__restrict__ sell_w,
+    const unsigned *__restrict__ currComm,
+    const double *__restrict__ vDegree,
+    const unsigned *__restrict__ sigma_inv, const Cinfo *__restrict__ cinfo,
+    Cinfo *__restrict__ cupd, double constant,
+    unsigned *__restrict__ targetComm, double *__restrict__ clusterWeight,
+    unsigned *__restrict__ spill_keys, double *__restrict__ spill_acc,
+    const i64 *__restrict__ spill_off) {
+    extern __shared__ __attribute__((aligned(16))) char smem[];
+    double *sacc = reinterpret_cast<double *>(smem);
+    unsigned *skey = reinterpret_cast<unsigned *>(
+        smem + sizeof(double) * SLOTS * blockDim.x);
+    const int tid = threadIdx.x;
+    const i64 gthread = blockIdx.x * (i64)blockDim.x + threadIdx.x;
+    const i64 stride = (i64)gridDim.x * blockDim.x;
+    unsigned *myspill_k = spill_keys + spill_off[gthread];
+    double *myspill_a = spill_acc + spill_off[gthread];
+
+    for (i64 s = s_begin + gthread; s < lnv; s += stride) {
+        const i64 i = perm[s];          // internal vertex index
+        const int deg = (int)deg_int[i];
+        const i64 ebase = chunk_off[s >> 6] + (s & 63);
+        const unsigned cc = currComm[i]; // label
+        const Cinfo ci = cinfo[sigma_inv[cc]];
+        const double ccDeg = ci.degree;
+        const i64 ccSize = ci.size;
+        unsigned target;
+        if (deg == 0) {
+            target = cc;          // dspl.hpp:323-324
+            clusterWeight[i] = 0; // K5 semantics (dspl.hpp:481-482)
+        } else {
+            double c0 = 0.0, selfLoop = 0.0;
+            int ns = 0, nspill = 0;
+            constexpr int CH = 8;
+            for (int k0 = 0; k0 < deg; k0 += CH) {
+                const int m = min(CH, deg - k0);
+                i64 tb[CH];
+                unsigned cb[CH];
+                double wb[CH];
+#pragma unroll
+                for (int j = 0; j < CH; j++) {
+                    const i64 slot =
+                        (j < m) ? ebase + (i64)(k0 + j) * 64 : ebase;
+                    tb[j] = sell_tidx[slot];
+                    if (!UNIT) wb[j] = sell_w[slot];
+                }
+#pragma unroll
+                for (int j = 0; j < CH; j++) cb[j] = currComm[tb[j]];
+                for (int j = 0; j < m; j++) {
+                    const i64 tidx = tb[j];
+                    const double w = UNIT ? 1.0 : wb[j];
+                    if (tidx == i) selfLoop += w; // dspl.hpp:247-248
+                    const unsigned tcomm = cb[j];
+                    if (tcomm == cc) { c0 += w; continue; }
+                    bool found = false;
+                    for (int t = 0; t < ns; t++) {
+                        if (skey[t * blockDim.x + tid] == tcomm) {
+                            sacc[t * blockDim.x + tid] += w;
+                            found = true;
+                            break;
+                        }
+                    }
+                    if (found) continue;
+                    if (ns < SLOTS) {
+                        skey[ns * blockDim.x + tid] = tcomm;
+                        sacc[ns * blockDim.x + tid] = w;
+                        ns++;
+                        continue;
+                    }
+                    for (int t = 0; t < nspill; t++) {
+                        if (myspill_k[t] == tcomm) {
+                            myspill_a[t] += w;
+                            found = true;
+                            break;
+                        }
+                    }
+                    if (!found) {
+                        myspill_k[nspill] = tcomm;
+                        myspill_a[nspill] = w;
+                        nspill++;
+                    }
+                }
+            }
+            clusterWeight[i] = c0; // dspl.hpp:318
+
+            // distGetMaxIndex (dspl.hpp:174-228)
+            const double vdeg = vDegree[i];
+            const double eix = c0 - selfLoop;
+            const double ax = ccDeg - vdeg;
+            double maxGain = 0.0;
+            unsigned maxIndex = cc;
+            i64 maxSize = ccSize;
+            const int tot = ns + nspill;
+            for (int t = 0; t < tot; t++) {
+                const unsigned y = (t < ns) ? skey[t * blockDim.x + tid]
+                                            : myspill_k[t - ns];
+                const double eiy = (t < ns) ? sacc[t * blockDim.x + tid]
+                                            : myspill_a[t - ns];
+                const Cinfo c = cinfo[sigma_inv[y]];
+                const double ay = c.degree;
+                const i64 ysz = c.size;
+                const double curGain =
+                    2.0 * (eiy - eix) - 2.0 * vdeg * (ay - ax) * constant; // :212
+                if (curGain > maxGain ||
+                    (curGain == maxGain && curGain != 0.0 && y < maxIndex)) {
+                    maxGain = curGain;
+                    maxIndex = y;
+                    maxSize = ysz;
+                }
+            }
+            if (maxSize == 1 && ccSize == 1 && maxIndex > cc) // :224-225
+                maxIndex = cc;
+            target = maxIndex;
+        }
+
+        if (target != cc) { // both communities local (dspl.hpp:331-399)
+            const double vdeg = vDegree[i];
+            Cinfo *u = &cupd[sigma_inv[cc]];
+            atomicAdd(&u->degree, -vdeg);
+            atomic_add_i64(&u->size, -1);
+            Cinfo *t = &cupd[sigma_inv[target]];
+            atomicAdd(&t->degree, vdeg);
+            atomic_add_i64(&t->size, 1);
+        }
+        targetComm[i] = target; // dspl.hpp:404 (label)
+    }
+}
+
+// iteration-1 streaming specialization, label mode (see k4_sweep_iter1's
+// header note; sigma_inv[cc] == i at iteration 1)
+template <bool UNIT>
+__global__ __launch_bounds__(256) void k4_sweep_iter1_p1(
+    i64 s_begin, i64 lnv, const unsigned *__restrict__ perm,
+    const unsigned *__restrict__ deg_int, const i64 *__restrict__ chunk_off,
+    const int *__restrict__ sell_tidx, const double *__restrict__ sell_w,
+    const unsigned *__restrict__ currComm,
+    const double *__restrict__ vDegree,
+    const unsigned *__restrict__ sigma_inv, Cinfo *__restrict__ cupd,
+    double constant, unsigned *__restrict__ targetComm,
+    double *__restrict__ clusterWeight) {
+    const i64 gthread = blockIdx.x * (i64)blockDim.x + threadIdx.x;
+    const i64 stride = (i64)gridDim.x * blockDim.x;
+    for (i64 s = s_begin + gthread; s < lnv; s += stride) {
+        const i64 i = perm[s];
+        const int deg = (int)deg_int[i];
+        const i64 ebase = chunk_off[s >> 6] + (s & 63);
+        const unsigned cc = currComm[i]; // own label
+        const double vdeg = vDegree[i];
+        double c0 = 0.0;
+        double maxGain = 0.0;
+        unsigned maxIndex = cc;
+        i64 prev = INT64_MIN;
+        unsigned pend_label = 0;
+        double eiy = 0.0, pend_ay = 0.0;
+        bool pend = false;
+        constexpr int CH = 8;
+        for (int k0 = 0; k0 < deg; k0 += CH) {
+            const int m = min(CH, deg - k0);
+            i64 tb[CH];
+            unsigned cb[CH];
+            double wb[CH], vb[CH];
+#pragma unroll
+            for (int j = 0; j < CH; j++) {
+                const i64 slot = (j < m) ? ebase + (i64)(k0 + j) * 64 : ebase;
+                tb[j] = sell_tidx[slot];
+                if (!UNIT) wb[j] = sell_w[slot];
+            }
+#pragma unroll
+            for (int j = 0; j < CH; j++) {
+                cb[j] = currComm[tb[j]];
+                vb[j] = vDegree[tb[j]];
+            }
+            for (int j = 0; j < m; j++) {
+                const i64 tidx = tb[j];
+                const double w = UNIT ? 1.0 : wb[j];
+                if (tidx == i) { c0 += w; continue; } // self: counter[0]
+                if (tidx == prev) { eiy += w; continue; } // parallel edge
+                if (pend) {
+                    const double g =
+                        2.0 * eiy - 2.0 * vdeg * pend_ay * constant;
+                    if (g > maxGain) { maxGain = g; maxIndex = pend_label; }
+                }
+                prev = tidx;
+                pend_label = cb[j]; // label-ordered like the tails
+                eiy = w;
+                pend_ay = vb[j];
+                pend = true;
+            }
+        }
+        if (pend) {
+            const double g = 2.0 * eiy - 2.0 * vdeg * pend_ay * constant;
+            if (g > maxGain) { maxGain = g; maxIndex = pend_label; }
+        }
+        if (maxIndex > cc) maxIndex = cc; // singleton guard
+        clusterWeight[i] = c0;            // dspl.hpp:318 (eix == 0)
+        if (maxIndex != cc) {
+            Cinfo *u = &cupd[i]; // sigma_inv[cc] == i at iteration 1
+            atomicAdd(&u->degree, -vdeg);
+            atomic_add_i64(&u->size, -1);
+            Cinfo *t = &cupd[sigma_inv[maxIndex]];
+            atomicAdd(&t->degree, vdeg);
+            atomic_add_i64(&t->size, 1);
+        }
+        targetComm[i] = maxIndex;
+    }
+}
+
+// wave-per-vertex hub path, label mode (see k4_sweep_hi's header note;
+// at p==1 a global tail id IS the local vertex id)
+__global__ __launch_bounds__(256) void k4_sweep_hi_p1(
+    i64 nhi, i64 lnv, const unsigned *__restrict__ perm,
+    const unsigned *__restrict__ deg_int, const unsigned *__restrict__ sigma,
+    const unsigned *__restrict__ sigma_inv, const i64 *__restrict__ xadj,
+    const i64 *__restrict__ tails, const unsigned *__restrict__ currComm,
+    const double *__restrict__ vDegree, const Cinfo *__restrict__ cinfo,
+    Cinfo *__restrict__ cupd, double constant,
+    unsigned *__restrict__ targetComm, double *__restrict__ clusterWeight,
+    const i64 *__restrict__ hash_off, i64 *__restrict__ hkeys,
+    double *__restrict__ hacc) {
+    const int lane = threadIdx.x & 63;
+    const int wid = threadIdx.x >> 6;
+    const int wpb = blockDim.x >> 6;
+    for (i64 s = (i64)blockIdx.x * wpb + wid; s < nhi;
+         s += (i64)gridDim.x * wpb) {
+        const i64 i = perm[s];
+        const i64 v = sigma[i];
+        const int deg = (int)deg_int[i];
+        const i64 e0 = xadj[v];
+        const unsigned cc = currComm[i]; // label
+        const i64 hoff = hash_off[s];
+        const i64 cap = hash_off[s + 1] - hoff; // power of two
+        const Cinfo cci = cinfo[sigma_inv[cc]];
+        const double ccDeg = cci.degree;
+        const i64 ccSize = cci.size;
+        double c0 = 0.0, selfLoop = 0.0;
+        for (int k = lane; k < deg; k += 64) {
+            const i64 tail = tails[e0 + k]; // global == local at p==1
+            const double w = 1.0;           // hub path is unit-only
+            if (tail == v) selfLoop += w;
+            const i64 tcomm = (i64)currComm[sigma_inv[tail]];
+            if (tcomm == (i64)cc) { c0 += w; continue; }
+            i64 pos = (i64)(((uint64_t)tcomm * 0x9E3779B97F4A7C15ull) >> 32) &
+                      (cap - 1);
+            for (;;) {
+                const i64 prev = (i64)atomicCAS(
+                    (unsigned long long *)&hkeys[hoff + pos],
+                    (unsigned long long)(-1ll), (unsigned long long)tcomm);
+                if (prev == -1 || prev == tcomm) {
+                    atomicAdd(&hacc[hoff + pos], w);
+                    break;
+                }
+                pos = (pos + 1) & (cap - 1);
+            }
+        }
+        for (int off = 32; off > 0; off >>= 1) {
+            c0 += __shfl_down(c0, off, 64);
+            selfLoop += __shfl_down(selfLoop, off, 64);
+        }
+        c0 = __shfl(c0, 0, 64);
+        selfLoop = __shfl(selfLoop, 0, 64);
+        const double vdeg = vDegree[i];
+        const double eix = c0 - selfLoop;
+        const double ax = ccDeg - vdeg;
+        double bg = 0.0;
+        i64 bl = INT64_MAX, bs = 0;
+        for (i64 t = lane; t < cap; t += 64) {
+            const i64 y = hkeys[hoff + t];
+            if (y == -1) continue;
+            const double eiy = hacc[hoff + t];
+            const Cinfo c = cinfo[sigma_inv[(unsigned)y]];
+            const double ay = c.degree;
+            const i64 ysz = c.size;
+            const double g =
+                2.0 * (eiy - eix) - 2.0 * vdeg * (ay - ax) * constant;
+            if (g > bg || (g == bg && g != 0.0 && y < bl)) {
+                bg = g;
+                bl = y;
+                bs = ysz;
+            }
+        }
+        for (int off = 32; off > 0; off >>= 1) {
+            const double og = __shfl_down(bg, off, 64);
+            const i64 ol = __shfl_down(bl, off, 64);
+            const i64 os = __shfl_down(bs, off, 64);
+            if (og > bg || (og == bg && ol < bl)) {
+                bg = og;
+                bl = ol;
+                bs = os;
+            }
+        }
+        if (lane == 0) {
+            i64 target = (bl == INT64_MAX) ? (i64)cc : bl;
+            if (bs == 1 && ccSize == 1 && target > (i64)cc) target = cc;
+            if (deg == 0) {
+                clusterWeight[i] = 0;
+                target = cc;
+            } else {
+                clusterWeight[i] = c0;
+            }
+            if (target != (i64)cc) {
+                Cinfo *u = &cupd[sigma_inv[cc]];
+                atomicAdd(&u->degree, -vdeg);
+                atomic_add_i64(&u->size, -1);
+                Cinfo *t = &cupd[sigma_inv[(unsigned)target]];
+                atomicAdd(&t->degree, vdeg);
+                atomic_add_i64(&t->size, 1);
+            }
+            targetComm[i] = (unsigned)target;
+        }
+    }
+}
+
 int grid_for(i64 n, int block = 256, int cap = 2048) {
     i64 g = (n + block - 1) / block;
     return (int)std::min<i64>(std::max<i64>(g, 1), cap);
@@ -1508,8 +1854,12 @@ double mv_engine_run(mv_engine *e, double lower, double thresh,
     }
     const double constant = 1.0 / totalW; // dspl.hpp:129
     PHASE("k2-done");
-    k3_init_comm<<<grid_for(lnv), 256, 0, st>>>(lnv, e->base, e->d_sigma,
-                                                e->d_curr, e->d_past);
+    if (p == 1) // label mode: the i64 arrays alias u32 label arrays
+        k3_init_comm32<<<grid_for(lnv), 256, 0, st>>>(
+            lnv, e->d_sigma, (unsigned *)e->d_curr, (unsigned *)e->d_past);
+    else
+        k3_init_comm<<<grid_for(lnv), 256, 0, st>>>(lnv, e->base, e->d_sigma,
+                                                    e->d_curr, e->d_past);
     HIP_CHECK(hipStreamSynchronize(st));
     PHASE("k3-done");
     e->stats.setup_ms =
@@ -1657,6 +2007,17 @@ double mv_engine_run(mv_engine *e, double lower, double thresh,
         const int slots = (numIters <= first_iters) ? slots_first : slots_rest;
         auto launch_sweep = [&](auto slots_tag, auto unit_tag) {
             constexpr int S = decltype(slots_tag)::value;
+            if (p == 1) { // u32 labels: half-size gathers, 12 B/lane LDS
+                k4_sweep_p1<S, decltype(unit_tag)::value>
+                    <<<e->sweep_grid, 256, S * 256 * 12, st>>>(
+                        e->nhi, lnv, e->d_perm, e->d_deg, e->d_chunk_off,
+                        e->d_sell_tidx, e->d_sell_w, (const unsigned *)d_curr,
+                        e->d_vdeg, e->d_sigma_inv, e->d_cinfo, e->d_cupd,
+                        constant, (unsigned *)d_target, e->d_cw,
+                        (unsigned *)e->d_spill_k, e->d_spill_a,
+                        e->d_spill_off);
+                return;
+            }
             k4_sweep<S, decltype(unit_tag)::value>
                 <<<e->sweep_grid, 256, S * 256 * 16, st>>>(
                     e->nhi, lnv, e->base, e->bound, e->d_perm, e->d_deg,
@@ -1677,6 +2038,15 @@ double mv_engine_run(mv_engine *e, double lower, double thresh,
             }
         };
         auto launch_iter1 = [&](auto unit_tag) {
+            if (p == 1) {
+                k4_sweep_iter1_p1<decltype(unit_tag)::value>
+                    <<<e->sweep_grid, 256, 0, st>>>(
+                        e->nhi, lnv, e->d_perm, e->d_deg, e->d_chunk_off,
+                        e->d_sell_tidx, e->d_sell_w, (const unsigned *)d_curr,
+                        e->d_vdeg, e->d_sigma_inv, e->d_cupd, constant,
+                        (unsigned *)d_target, e->d_cw);
+                return;
+            }
             k4_sweep_iter1<decltype(unit_tag)::value>
                 <<<e->sweep_grid, 256, 0, st>>>(
                     e->nhi, lnv, e->base, e->bound, e->d_perm, e->d_deg,
@@ -1687,6 +2057,15 @@ double mv_engine_run(mv_engine *e, double lower, double thresh,
         if (e->nhi > 0) { // wave-per-vertex hubs (unit weights only)
             HIP_CHECK(hipMemsetAsync(e->d_hkeys, 0xFF, 8 * e->hash_total, st));
             HIP_CHECK(hipMemsetAsync(e->d_hacc, 0, 8 * e->hash_total, st));
+            if (p == 1)
+                k4_sweep_hi_p1<<<grid_for(e->nhi * 64, 256, 2048), 256, 0,
+                                 st>>>(
+                    e->nhi, lnv, e->d_perm, e->d_deg, e->d_sigma,
+                    e->d_sigma_inv, e->d_xadj, e->d_tails,
+                    (const unsigned *)d_curr, e->d_vdeg, e->d_cinfo,
+                    e->d_cupd, constant, (unsigned *)d_target, e->d_cw,
+                    e->d_hash_off, e->d_hkeys, e->d_hacc);
+            else
             k4_sweep_hi<true><<<grid_for(e->nhi * 64, 256, 2048), 256, 0,
                                 st>>>(
                 e->nhi, lnv, e->base, e->bound, e->d_perm, e->d_deg,
@@ -1777,9 +2156,13 @@ double mv_engine_run(mv_engine *e, double lower, double thresh,
         if (e->trace_mod && numIters <= e->trace_cap)
             e->trace_mod[numIters - 1] = currMod;
         if (e->trace_target && numIters <= e->trace_cap) {
-            k_depermute<<<grid_for(lnv), 256, 0, st>>>(lnv, e->d_sigma_inv,
-                                                       d_target,
-                                                       e->d_trace_tmp);
+            if (p == 1)
+                k_depermute32<<<grid_for(lnv), 256, 0, st>>>(
+                    lnv, e->d_sigma_inv, (const unsigned *)d_target,
+                    e->d_trace_tmp);
+            else
+                k_depermute<<<grid_for(lnv), 256, 0, st>>>(
+                    lnv, e->d_sigma_inv, d_target, e->d_trace_tmp);
             HIP_CHECK(hipMemcpyAsync(e->trace_target +
                                          (i64)(numIters - 1) * lnv,
                                      e->d_trace_tmp, 8 * lnv,

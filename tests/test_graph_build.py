@@ -190,4 +190,8 @@ def test_balanced_read_matches_reference():
         og.free()
     g.free()
     assert iters == int(m.group(2))
-    assert abs(mod - float(m.group(1))) < 5e-7
+    # %g prints 6 significant digits (granularity 1e-6), and the reference
+    # binary itself shows rare run-to-run wobble of ~1e-6 here (observed
+    # 0.752137 vs the usual 0.752138 on identical input); a real read bug
+    # moves modularity by far more, and iters above stays strict.
+    assert abs(mod - float(m.group(1))) < 1.5e-6
