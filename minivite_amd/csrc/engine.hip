@@ -874,33 +874,34 @@ __global__ __launch_bounds__(256) void k4_sweep_iter1(
 }
 
 
-// ---- Single-rank (p==1) specializations: u32 LABEL communities ----
+// ---- Single-rank (p==1) specializations: u32 SLOT communities ----
 // At one rank every community is local (base 0, no ghosts, no remote
-// paths), so a community value can be its 32-bit LABEL instead of the
-// 64-bit handle: label comparisons keep the reference tie-break and
-// singleton-guard semantics bit-for-bit, and the per-edge community
-// gather — the sweep's dominant random traffic — halves to 4 B. Candidate
-// info stays in internal (spatial) order; the label->slot hop goes
-// through sigma_inv once per DISTINCT candidate, not per edge. Same
-// reference anchors as the generic kernels (dspl.hpp:174-405).
-__global__ void k3_init_comm32(i64 lnv, const unsigned *__restrict__ sigma,
-                               unsigned *__restrict__ curr,
+// paths), so a community value can be the 32-bit internal SLOT of its
+// home vertex: equality probes only need community identity, and
+// cinfo/cupd indexing needs exactly the slot — so the per-edge community
+// gather (the sweep's dominant random traffic) halves to 4 B with NO
+// added indirection on the info lookups. Only the reference's tie-break
+// and singleton guard need label ORDER (dspl.hpp:214-225), and only per
+// DISTINCT candidate: the label is sigma[slot], an independent
+// spatially-clustered gather. (A pure-label variant was measured and
+// rejected: sigma_inv[label]->cinfo adds a dependent-load chain to the
+// latency-bound gain scan — 4M 104G vs 110G handles vs this design.)
+__global__ void k3_init_comm32(i64 lnv, unsigned *__restrict__ curr,
                                unsigned *__restrict__ past) {
     for (i64 k = blockIdx.x * (i64)blockDim.x + threadIdx.x; k < lnv;
          k += (i64)gridDim.x * blockDim.x) {
-        const unsigned l = sigma[k]; // label of the vertex slot k holds
-        curr[k] = l;
-        past[k] = l;
+        curr[k] = (unsigned)k; // slot k's own community (dspl.hpp:132-149)
+        past[k] = (unsigned)k;
     }
 }
 
-__global__ void k_depermute32(i64 lnv,
+__global__ void k_depermute32(i64 lnv, const unsigned *__restrict__ sigma,
                               const unsigned *__restrict__ sigma_inv,
                               const unsigned *__restrict__ in,
                               i64 *__restrict__ out) {
     for (i64 v = blockIdx.x * (i64)blockDim.x + threadIdx.x; v < lnv;
          v += (i64)gridDim.x * blockDim.x)
-        out[v] = (i64)in[sigma_inv[v]];
+        out[v] = (i64)sigma[in[sigma_inv[v]]]; // slot -> label
 }
 
 template <int SLOTS, bool UNIT>
@@ -909,12 +910,11 @@ __global__ __launch_bounds__(256) void k4_sweep_p1(
     const unsigned *__restrict__ deg_int, const i64 *__restrict__ chunk_off,
     const int *__restrict__ sell_tidx, const double *__restrict__ sell_w,
     const unsigned *__restrict__ currComm,
-    const double *__restrict__ vDegree,
-    const unsigned *__restrict__ sigma_inv, const Cinfo *__restrict__ cinfo,
-    Cinfo *__restrict__ cupd, double constant,
-    unsigned *__restrict__ targetComm, double *__restrict__ clusterWeight,
-    unsigned *__restrict__ spill_keys, double *__restrict__ spill_acc,
-    const i64 *__restrict__ spill_off) {
+    const double *__restrict__ vDegree, const unsigned *__restrict__ sigma,
+    const Cinfo *__restrict__ cinfo, Cinfo *__restrict__ cupd,
+    double constant, unsigned *__restrict__ targetComm,
+    double *__restrict__ clusterWeight, unsigned *__restrict__ spill_keys,
+    double *__restrict__ spill_acc, const i64 *__restrict__ spill_off) {
     extern __shared__ __attribute__((aligned(16))) char smem[];
     double *sacc = reinterpret_cast<double *>(smem);
     unsigned *skey = reinterpret_cast<unsigned *>(
@@ -929,8 +929,8 @@ __global__ __launch_bounds__(256) void k4_sweep_p1(
         const i64 i = perm[s];          // internal vertex index
         const int deg = (int)deg_int[i];
         const i64 ebase = chunk_off[s >> 6] + (s & 63);
-        const unsigned cc = currComm[i]; // label
-        const Cinfo ci = cinfo[sigma_inv[cc]];
+        const unsigned cc = currComm[i]; // slot
+        const Cinfo ci = cinfo[cc];
         const double ccDeg = ci.degree;
         const i64 ccSize = ci.size;
         unsigned target;
@@ -992,7 +992,10 @@ __global__ __launch_bounds__(256) void k4_sweep_p1(
             }
             clusterWeight[i] = c0; // dspl.hpp:318
 
-            // distGetMaxIndex (dspl.hpp:174-228)
+            // distGetMaxIndex (dspl.hpp:174-228); order on LABELS, but
+            // labels (sigma[slot]) are fetched LAZILY — only when a gain
+            // tie actually needs the order — so the common strict-greater
+            // path costs no label traffic at all.
             const double vdeg = vDegree[i];
             const double eix = c0 - selfLoop;
             const double ax = ccDeg - vdeg;
@@ -1005,68 +1008,77 @@ __global__ __launch_bounds__(256) void k4_sweep_p1(
                                             : myspill_k[t - ns];
                 const double eiy = (t < ns) ? sacc[t * blockDim.x + tid]
                                             : myspill_a[t - ns];
-                const Cinfo c = cinfo[sigma_inv[y]];
+                const Cinfo c = cinfo[y];
                 const double ay = c.degree;
                 const i64 ysz = c.size;
                 const double curGain =
                     2.0 * (eiy - eix) - 2.0 * vdeg * (ay - ax) * constant; // :212
-                if (curGain > maxGain ||
-                    (curGain == maxGain && curGain != 0.0 && y < maxIndex)) {
+                if (curGain > maxGain) {
                     maxGain = curGain;
                     maxIndex = y;
                     maxSize = ysz;
+                } else if (curGain == maxGain && curGain != 0.0) {
+                    // real branch, not a select: the label loads must not
+                    // be folded into the gain-compare dataflow
+                    if (sigma[y] < sigma[maxIndex]) {
+                        maxIndex = y;
+                        maxSize = ysz;
+                    }
                 }
             }
-            if (maxSize == 1 && ccSize == 1 && maxIndex > cc) // :224-225
-                maxIndex = cc;
+            if (maxSize == 1 && ccSize == 1 && maxIndex != cc) { // :224-225
+                if (sigma[maxIndex] > sigma[cc]) maxIndex = cc;
+            }
             target = maxIndex;
         }
 
         if (target != cc) { // both communities local (dspl.hpp:331-399)
             const double vdeg = vDegree[i];
-            Cinfo *u = &cupd[sigma_inv[cc]];
+            Cinfo *u = &cupd[cc];
             atomicAdd(&u->degree, -vdeg);
             atomic_add_i64(&u->size, -1);
-            Cinfo *t = &cupd[sigma_inv[target]];
+            Cinfo *t = &cupd[target];
             atomicAdd(&t->degree, vdeg);
             atomic_add_i64(&t->size, 1);
         }
-        targetComm[i] = target; // dspl.hpp:404 (label)
+        targetComm[i] = target; // dspl.hpp:404 (slot)
     }
 }
 
-// iteration-1 streaming specialization, label mode (see k4_sweep_iter1's
-// header note; sigma_inv[cc] == i at iteration 1)
+// iteration-1 streaming specialization, slot mode (see k4_sweep_iter1's
+// header note). currComm is the identity permutation, so a candidate
+// community IS its tail index — no community gather at all; the
+// label-order stream property and ay = vDegree[tidx] carry over.
 template <bool UNIT>
 __global__ __launch_bounds__(256) void k4_sweep_iter1_p1(
     i64 s_begin, i64 lnv, const unsigned *__restrict__ perm,
     const unsigned *__restrict__ deg_int, const i64 *__restrict__ chunk_off,
     const int *__restrict__ sell_tidx, const double *__restrict__ sell_w,
-    const unsigned *__restrict__ currComm,
-    const double *__restrict__ vDegree,
-    const unsigned *__restrict__ sigma_inv, Cinfo *__restrict__ cupd,
-    double constant, unsigned *__restrict__ targetComm,
-    double *__restrict__ clusterWeight) {
+    const double *__restrict__ vDegree, const unsigned *__restrict__ sigma,
+    Cinfo *__restrict__ cupd, double constant,
+    unsigned *__restrict__ targetComm, double *__restrict__ clusterWeight) {
     const i64 gthread = blockIdx.x * (i64)blockDim.x + threadIdx.x;
     const i64 stride = (i64)gridDim.x * blockDim.x;
     for (i64 s = s_begin + gthread; s < lnv; s += stride) {
         const i64 i = perm[s];
         const int deg = (int)deg_int[i];
         const i64 ebase = chunk_off[s >> 6] + (s & 63);
-        const unsigned cc = currComm[i]; // own label
+        const unsigned cc = (unsigned)i; // own slot
+        const unsigned ccLabel = sigma[i];
         const double vdeg = vDegree[i];
         double c0 = 0.0;
         double maxGain = 0.0;
         unsigned maxIndex = cc;
+        unsigned maxLabel = ccLabel;
         i64 prev = INT64_MIN;
-        unsigned pend_label = 0;
+        unsigned pend_slot = 0, pend_label = 0;
         double eiy = 0.0, pend_ay = 0.0;
         bool pend = false;
         constexpr int CH = 8;
         for (int k0 = 0; k0 < deg; k0 += CH) {
             const int m = min(CH, deg - k0);
             i64 tb[CH];
-            unsigned cb[CH];
+            unsigned lb[CH];
             double wb[CH], vb[CH];
 #pragma unroll
             for (int j = 0; j < CH; j++) {
@@ -1076,7 +1088,7 @@ __global__ __launch_bounds__(256) void k4_sweep_iter1_p1(
             }
 #pragma unroll
             for (int j = 0; j < CH; j++) {
-                cb[j] = currComm[tb[j]];
+                lb[j] = sigma[tb[j]];
                 vb[j] = vDegree[tb[j]];
             }
             for (int j = 0; j < m; j++) {
@@ -1087,10 +1099,15 @@ __global__ __launch_bounds__(256) void k4_sweep_iter1_p1(
                 if (pend) {
                     const double g =
                         2.0 * eiy - 2.0 * vdeg * pend_ay * constant;
-                    if (g > maxGain) { maxGain = g; maxIndex = pend_label; }
+                    if (g > maxGain) {
+                        maxGain = g;
+                        maxIndex = pend_slot;
+                        maxLabel = pend_label;
+                    }
                 }
                 prev = tidx;
-                pend_label = cb[j]; // label-ordered like the tails
+                pend_slot = (unsigned)tidx; // candidate community == tail
+                pend_label = lb[j];         // label-ordered like the tails
                 eiy = w;
                 pend_ay = vb[j];
                 pend = true;
@@ -1098,15 +1115,19 @@ __global__ __launch_bounds__(256) void k4_sweep_iter1_p1(
         }
         if (pend) {
             const double g = 2.0 * eiy - 2.0 * vdeg * pend_ay * constant;
-            if (g > maxGain) { maxGain = g; maxIndex = pend_label; }
+            if (g > maxGain) {
+                maxGain = g;
+                maxIndex = pend_slot;
+                maxLabel = pend_label;
+            }
         }
-        if (maxIndex > cc) maxIndex = cc; // singleton guard
-        clusterWeight[i] = c0;            // dspl.hpp:318 (eix == 0)
+        if (maxLabel > ccLabel) maxIndex = cc; // singleton guard
+        clusterWeight[i] = c0;                 // dspl.hpp:318 (eix == 0)
         if (maxIndex != cc) {
-            Cinfo *u = &cupd[i]; // sigma_inv[cc] == i at iteration 1
+            Cinfo *u = &cupd[i];
             atomicAdd(&u->degree, -vdeg);
             atomic_add_i64(&u->size, -1);
-            Cinfo *t = &cupd[sigma_inv[maxIndex]];
+            Cinfo *t = &cupd[maxIndex];
             atomicAdd(&t->degree, vdeg);
             atomic_add_i64(&t->size, 1);
         }
@@ -1114,8 +1135,10 @@ __global__ __launch_bounds__(256) void k4_sweep_iter1_p1(
     }
 }
 
-// wave-per-vertex hub path, label mode (see k4_sweep_hi's header note;
-// at p==1 a global tail id IS the local vertex id)
+// wave-per-vertex hub path, slot mode (see k4_sweep_hi's header note;
+// at p==1 a global tail id IS the local vertex id). The wave-reduce
+// tie-break order rides a packed (label << 32 | slot) value, so label
+// order decides and the slot comes along for free.
 __global__ __launch_bounds__(256) void k4_sweep_hi_p1(
     i64 nhi, i64 lnv, const unsigned *__restrict__ perm,
     const unsigned *__restrict__ deg_int, const unsigned *__restrict__ sigma,
@@ -1135,10 +1158,11 @@ __global__ __launch_bounds__(256) void k4_sweep_hi_p1(
         const i64 v = sigma[i];
         const int deg = (int)deg_int[i];
         const i64 e0 = xadj[v];
-        const unsigned cc = currComm[i]; // label
+        const unsigned cc = currComm[i]; // slot
+        const unsigned ccLabel = sigma[cc];
         const i64 hoff = hash_off[s];
         const i64 cap = hash_off[s + 1] - hoff; // power of two
-        const Cinfo cci = cinfo[sigma_inv[cc]];
+        const Cinfo cci = cinfo[cc];
         const double ccDeg = cci.degree;
         const i64 ccSize = cci.size;
         double c0 = 0.0, selfLoop = 0.0;
@@ -1171,19 +1195,20 @@ __global__ __launch_bounds__(256) void k4_sweep_hi_p1(
         const double eix = c0 - selfLoop;
         const double ax = ccDeg - vdeg;
         double bg = 0.0;
-        i64 bl = INT64_MAX, bs = 0;
+        i64 bl = INT64_MAX, bs = 0; // bl = (label << 32) | slot
         for (i64 t = lane; t < cap; t += 64) {
             const i64 y = hkeys[hoff + t];
             if (y == -1) continue;
             const double eiy = hacc[hoff + t];
-            const Cinfo c = cinfo[sigma_inv[(unsigned)y]];
+            const Cinfo c = cinfo[y];
             const double ay = c.degree;
             const i64 ysz = c.size;
             const double g =
                 2.0 * (eiy - eix) - 2.0 * vdeg * (ay - ax) * constant;
-            if (g > bg || (g == bg && g != 0.0 && y < bl)) {
+            const i64 yh = ((i64)sigma[y] << 32) | y;
+            if (g > bg || (g == bg && g != 0.0 && yh < bl)) {
                 bg = g;
-                bl = y;
+                bl = yh;
                 bs = ysz;
             }
         }
@@ -1198,23 +1223,27 @@ __global__ __launch_bounds__(256) void k4_sweep_hi_p1(
             }
         }
         if (lane == 0) {
-            i64 target = (bl == INT64_MAX) ? (i64)cc : bl;
-            if (bs == 1 && ccSize == 1 && target > (i64)cc) target = cc;
+            unsigned target =
+                (bl == INT64_MAX) ? cc : (unsigned)(bl & 0xffffffffu);
+            const unsigned tLabel =
+                (bl == INT64_MAX) ? ccLabel : (unsigned)(bl >> 32);
+            if (bs == 1 && ccSize == 1 && tLabel > ccLabel)
+                target = cc; // :224-225
             if (deg == 0) {
                 clusterWeight[i] = 0;
                 target = cc;
             } else {
                 clusterWeight[i] = c0;
             }
-            if (target != (i64)cc) {
-                Cinfo *u = &cupd[sigma_inv[cc]];
+            if (target != cc) {
+                Cinfo *u = &cupd[cc];
                 atomicAdd(&u->degree, -vdeg);
                 atomic_add_i64(&u->size, -1);
-                Cinfo *t = &cupd[sigma_inv[(unsigned)target]];
+                Cinfo *t = &cupd[target];
                 atomicAdd(&t->degree, vdeg);
                 atomic_add_i64(&t->size, 1);
             }
-            targetComm[i] = (unsigned)target;
+            targetComm[i] = target;
         }
     }
 }
@@ -1854,9 +1883,9 @@ double mv_engine_run(mv_engine *e, double lower, double thresh,
     }
     const double constant = 1.0 / totalW; // dspl.hpp:129
     PHASE("k2-done");
-    if (p == 1) // label mode: the i64 arrays alias u32 label arrays
+    if (p == 1) // slot mode: the i64 arrays alias u32 slot arrays
         k3_init_comm32<<<grid_for(lnv), 256, 0, st>>>(
-            lnv, e->d_sigma, (unsigned *)e->d_curr, (unsigned *)e->d_past);
+            lnv, (unsigned *)e->d_curr, (unsigned *)e->d_past);
     else
         k3_init_comm<<<grid_for(lnv), 256, 0, st>>>(lnv, e->base, e->d_sigma,
                                                     e->d_curr, e->d_past);
@@ -2007,12 +2036,12 @@ double mv_engine_run(mv_engine *e, double lower, double thresh,
         const int slots = (numIters <= first_iters) ? slots_first : slots_rest;
         auto launch_sweep = [&](auto slots_tag, auto unit_tag) {
             constexpr int S = decltype(slots_tag)::value;
-            if (p == 1) { // u32 labels: half-size gathers, 12 B/lane LDS
+            if (p == 1) { // u32 slots: half-size gathers, 12 B/lane LDS
                 k4_sweep_p1<S, decltype(unit_tag)::value>
                     <<<e->sweep_grid, 256, S * 256 * 12, st>>>(
                         e->nhi, lnv, e->d_perm, e->d_deg, e->d_chunk_off,
                         e->d_sell_tidx, e->d_sell_w, (const unsigned *)d_curr,
-                        e->d_vdeg, e->d_sigma_inv, e->d_cinfo, e->d_cupd,
+                        e->d_vdeg, e->d_sigma, e->d_cinfo, e->d_cupd,
                         constant, (unsigned *)d_target, e->d_cw,
                         (unsigned *)e->d_spill_k, e->d_spill_a,
                         e->d_spill_off);
@@ -2042,9 +2071,8 @@ double mv_engine_run(mv_engine *e, double lower, double thresh,
                 k4_sweep_iter1_p1<decltype(unit_tag)::value>
                     <<<e->sweep_grid, 256, 0, st>>>(
                         e->nhi, lnv, e->d_perm, e->d_deg, e->d_chunk_off,
-                        e->d_sell_tidx, e->d_sell_w, (const unsigned *)d_curr,
-                        e->d_vdeg, e->d_sigma_inv, e->d_cupd, constant,
-                        (unsigned *)d_target, e->d_cw);
+                        e->d_sell_tidx, e->d_sell_w, e->d_vdeg, e->d_sigma,
+                        e->d_cupd, constant, (unsigned *)d_target, e->d_cw);
                 return;
             }
             k4_sweep_iter1<decltype(unit_tag)::value>
@@ -2158,8 +2186,8 @@ double mv_engine_run(mv_engine *e, double lower, double thresh,
         if (e->trace_target && numIters <= e->trace_cap) {
             if (p == 1)
                 k_depermute32<<<grid_for(lnv), 256, 0, st>>>(
-                    lnv, e->d_sigma_inv, (const unsigned *)d_target,
-                    e->d_trace_tmp);
+                    lnv, e->d_sigma, e->d_sigma_inv,
+                    (const unsigned *)d_target, e->d_trace_tmp);
             else
                 k_depermute<<<grid_for(lnv), 256, 0, st>>>(
                     lnv, e->d_sigma_inv, d_target, e->d_trace_tmp);
