@@ -242,7 +242,7 @@ def _rank_main(rank, world, nv, out_q):
     td.destroy_process_group()
 
 
-@pytest.mark.parametrize("nv,world", [(16384, 2), (16384, 4)])
+@pytest.mark.parametrize("nv,world", [(16384, 2), (16384, 4), (32768, 8)])
 def test_gloo_protocol_matches_pin(nv, world):
     import torch.multiprocessing as mp
     pins = json.load(open(GOLDEN))
