@@ -21,7 +21,9 @@ from oracle.oracle import OracleGraph
     (16384, 8, True),
     (16384, 16, True),
     (32768, 8, True),
+    (65536, 32, True),
     (16384, 2, False),
+    (32768, 4, False),
 ])
 def test_rgg_bit_identical_to_oracle(nv, p, unit):
     og = OracleGraph.rgg(nv, p, unit_weight=unit)
