@@ -1440,6 +1440,11 @@ static void free_graph_state(mv_engine *e) {
         HIP_CHECK(hipHostFree(e->h_partials));
         e->h_partials = nullptr;
     }
+    // a stale trace against a freed d_trace_tmp (or a new lnv) would fault:
+    // re-arm via mv_engine_set_trace after every load
+    e->trace_target = nullptr;
+    e->trace_mod = nullptr;
+    e->trace_cap = 0;
     e->sell_elems = 0;
     e->spill_elems = 0;
     e->rc_cap = 0;
@@ -2142,9 +2147,17 @@ double mv_engine_run(mv_engine *e, double lower, double thresh,
             const auto t_h0 = std::chrono::steady_clock::now();
             rccl_alltoallv(e, e->d_rcu, rc_bounds.data(), e->d_req_info,
                            req_off.data(), sizeof(Info16), ncclChar, 1);
-            k_apply_deltas<<<grid_for(std::max<i64>(req_off[p], 1)), 256, 0,
-                             st>>>(req_off[p], e->d_req_ids, e->base,
-                                   e->d_sigma_inv, e->d_req_info, e->d_cinfo);
+            // apply per SENDER segment, in rank order: same-stream launches
+            // serialize, and one sender's ids are unique (sorted rc_ids), so
+            // cinfo degree bits are run-to-run identical at any nranks —
+            // like the reference's in-order delta walk (dspl.hpp:1089-1102)
+            for (int r = 0; r < p; r++) {
+                const i64 c = req_off[r + 1] - req_off[r];
+                if (c > 0)
+                    k_apply_deltas<<<grid_for(c), 256, 0, st>>>(
+                        c, e->d_req_ids + req_off[r], e->base, e->d_sigma_inv,
+                        e->d_req_info + req_off[r], e->d_cinfo);
+            }
             HIP_CHECK(hipStreamSynchronize(st));
             e->stats.halo_ms +=
                 std::chrono::duration<double, std::milli>(

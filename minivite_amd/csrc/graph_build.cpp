@@ -398,11 +398,17 @@ mv_graph *mv_graph_read_binary(const char *path, int rank, int nranks,
         std::vector<int64_t> mbins(nranks + 1, 0), nbins(nranks, 0);
         const int64_t nbcap = ne / nranks;
         int p = 0;
+        // The reference scans off[0..nv-1] sequentially and bins vertex m
+        // by off[m] - off[m-1] (with off[-1] = 0) — i.e. the degree LAGGED
+        // by one vertex (graph.hpp:437-451). Reproduced exactly: the -b
+        // partition must match the reference's bit-for-bit.
+        int64_t past = 0;
         for (int64_t m = 0; m < nv; m++) {
-            const int64_t deg = off[m + 1] - off[m];
+            const int64_t deg = off[m] - past;
             if (nbins[p] < nbcap || p == nranks - 1) nbins[p] += deg;
             if (nbins[p] >= nbcap && p < nranks - 1) p++;
             mbins[p + 1]++;
+            past = off[m];
         }
         for (int k = 1; k <= nranks; k++) mbins[k] += mbins[k - 1];
         g->parts = mbins; // repart, graph.hpp:511
@@ -410,8 +416,8 @@ mv_graph *mv_graph_read_binary(const char *path, int rank, int nranks,
     const int64_t v0 = g->parts[rank], v1 = g->parts[rank + 1];
     const int64_t lnv = v1 - v0;
     g->xadj.resize(lnv + 1);
-    // offsets slice
-    std::fseek(f, (long)(16 + v0 * 8), SEEK_SET);
+    // offsets slice (fseeko: offsets exceed 2^31 for >2 GB files)
+    fseeko(f, (off_t)(16 + v0 * 8), SEEK_SET);
     if (std::fread(g->xadj.data(), 8, lnv + 1, f) != (size_t)(lnv + 1)) {
         std::fclose(f);
         delete g;
@@ -424,7 +430,7 @@ mv_graph *mv_graph_read_binary(const char *path, int rank, int nranks,
     g->weights.resize(lne);
     std::vector<char> buf(lne * 16);
     const int64_t edge_base = 16 + (nv + 1) * 8;
-    std::fseek(f, (long)(edge_base + e0 * 16), SEEK_SET);
+    fseeko(f, (off_t)(edge_base + e0 * 16), SEEK_SET);
     if (lne && std::fread(buf.data(), 16, lne, f) != (size_t)lne) {
         std::fclose(f);
         delete g;

@@ -94,6 +94,46 @@ def test_balanced_read_partition():
         g.free()
 
 
+@pytest.mark.parametrize("p", [2, 4, 8])
+def test_balanced_partition_matches_reference_scan(p):
+    """-b parts bit-identical to find_balanced_num_edges (graph.hpp:437-456):
+    the reference bins vertex m by off[m] - off[m-1] (the LAGGED degree,
+    off[-1] = 0), not by its own degree. Restated here directly from the
+    file bytes and compared with mv_graph_read_binary's partition."""
+    import ctypes
+    from minivite_amd import lib
+    g = Graph.rgg(16384, 0, 1)
+    try:
+        with tempfile.TemporaryDirectory() as d:
+            path = os.path.join(d, "g.bin")
+            g.write_binary(path)
+            raw = np.fromfile(path, dtype=np.int64, count=2 + 16384)
+            nv, ne = int(raw[0]), int(raw[1])
+            off = raw[2:2 + nv]  # the reference reads nv offsets (off[0..nv-1])
+            nbcap = ne // p
+            nbins = [0] * p
+            mbins = [0] * (p + 1)
+            bp = 0
+            past = 0
+            for m in range(nv):
+                delta = int(off[m]) - past
+                if nbins[bp] < nbcap or bp == p - 1:
+                    nbins[bp] += delta
+                if nbins[bp] >= nbcap and bp < p - 1:
+                    bp += 1
+                mbins[bp + 1] += 1
+                past = int(off[m])
+            for k in range(1, p + 1):
+                mbins[k] += mbins[k - 1]
+            gr = Graph.read_binary(path, 0, p, balanced=True)
+            pp = lib().mv_graph_parts(gr.h)
+            parts = [pp[k] for k in range(p + 1)]
+            gr.free()
+            assert parts == mbins
+    finally:
+        g.free()
+
+
 def test_oracle_consumes_product_graph():
     """Louvain over a product-built (from_csr path) graph matches the pin —
     the same-graph guarantee behind every CPU-baseline comparison."""
