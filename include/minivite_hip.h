@@ -91,6 +91,20 @@ mv_engine *mv_engine_create(int device, int rank, int nranks,
                             const void *comm_id_or_null);
 void mv_engine_destroy(mv_engine *e);
 
+/* ---- loopback transport (hardware-validation harness) ----
+ * Runs nranks engines in ONE process on ONE device, one host thread per
+ * rank; the collectives become device-to-device copies + host barriers
+ * while every kernel and every byte of the multi-rank protocol stays the
+ * p>1 production path. Purpose: executing the partitioned engine on a
+ * single-GPU box, where RCCL refuses two ranks on one device. The product
+ * multi-GPU path (bench.py, mv355) always uses RCCL. */
+typedef struct mv_lb_session mv_lb_session;
+mv_lb_session *mv_lb_create(int nranks);
+void mv_lb_destroy(mv_lb_session *s);
+/* Like mv_engine_create but exchanges ride the loopback session. */
+mv_engine *mv_engine_create_lb(int device, int rank, int nranks,
+                               mv_lb_session *s);
+
 /* Upload the per-rank CSR to HBM (device layout: DESIGN.md §data layout). */
 int mv_engine_load_graph(mv_engine *e, const mv_graph *g);
 

@@ -8,6 +8,7 @@ engine without a GPU fails loudly.
 from .api import (  # noqa: F401
     Graph,
     Engine,
+    LoopbackSession,
     comm_id,
     lib,
 )

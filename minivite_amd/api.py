@@ -61,6 +61,13 @@ def lib():
         L.mv_graph_weights.argtypes = [vp]
         L.mv_comm_id.restype = ctypes.c_int
         L.mv_comm_id.argtypes = [ctypes.c_void_p]
+        L.mv_lb_create.restype = vp
+        L.mv_lb_create.argtypes = [ctypes.c_int]
+        L.mv_lb_destroy.restype = None
+        L.mv_lb_destroy.argtypes = [vp]
+        L.mv_engine_create_lb.restype = vp
+        L.mv_engine_create_lb.argtypes = [ctypes.c_int, ctypes.c_int,
+                                          ctypes.c_int, vp]
         L.mv_engine_create.restype = vp
         L.mv_engine_create.argtypes = [ctypes.c_int, ctypes.c_int, ctypes.c_int,
                                        ctypes.c_void_p]
@@ -163,6 +170,28 @@ class Graph:
             pass
 
 
+class LoopbackSession:
+    """Hardware-validation harness: nranks engines in ONE process on ONE
+    device (see minivite_hip.h). Test-only; product multi-GPU is RCCL."""
+
+    def __init__(self, nranks):
+        self.nranks = nranks
+        self.h = lib().mv_lb_create(nranks)
+        if not self.h:
+            raise RuntimeError("mv_lb_create failed")
+
+    def destroy(self):
+        if getattr(self, "h", None):
+            lib().mv_lb_destroy(self.h)
+            self.h = None
+
+    def __del__(self):
+        try:
+            self.destroy()
+        except Exception:
+            pass
+
+
 class Engine:
     """The GPU Louvain engine (one per process, one GPU)."""
 
@@ -176,6 +205,17 @@ class Engine:
                 "mv_engine_create failed — an MI355X GPU is required; "
                 "there is no CPU fallback")
         self._trace_buf = None
+
+    @classmethod
+    def loopback(cls, session, rank, device=0):
+        """Rank `rank` of a LoopbackSession (all ranks share `device`)."""
+        obj = cls.__new__(cls)
+        obj.h = lib().mv_engine_create_lb(device, rank, session.nranks,
+                                          session.h)
+        if not obj.h:
+            raise RuntimeError("mv_engine_create_lb failed")
+        obj._trace_buf = None
+        return obj
 
     def load_graph(self, g):
         if lib().mv_engine_load_graph(self.h, g.h) != 0:

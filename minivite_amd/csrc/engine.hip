@@ -44,10 +44,12 @@
 
 #include <algorithm>
 #include <chrono>
+#include <condition_variable>
 #include <cstdint>
 #include <cstdio>
 #include <cstdlib>
 #include <cstring>
+#include <mutex>
 #include <type_traits>
 #include <vector>
 
@@ -1256,9 +1258,44 @@ int grid_for(i64 n, int block = 256, int cap = 2048) {
 } // namespace
 
 // ------------------------------------------------------------------------
+// Loopback transport: nranks engines in one process on one device, one
+// host thread per rank (see the header note in minivite_hip.h). Exchange
+// payloads move by device-to-device hipMemcpyAsync between the engines'
+// buffers; host barriers stand in for the collective rendezvous. All
+// reductions run in rank order (deterministic, like the RCCL fixed-tree
+// at these tiny sizes and like the oracle).
+struct mv_lb_session {
+    int nranks;
+    std::mutex mu;
+    std::condition_variable cv;
+    int count = 0;
+    uint64_t gen = 0;
+    struct Post {
+        const void *send = nullptr;
+        const int64_t *soff = nullptr;
+        size_t eb = 0;
+        const double *red = nullptr;
+        std::vector<int64_t> cnts;
+    };
+    std::vector<Post> posts;
+    explicit mv_lb_session(int n) : nranks(n), posts(n) {}
+    void barrier() {
+        std::unique_lock<std::mutex> lk(mu);
+        const uint64_t g = gen;
+        if (++count == nranks) {
+            count = 0;
+            gen++;
+            cv.notify_all();
+        } else {
+            cv.wait(lk, [&] { return gen != g; });
+        }
+    }
+};
+
 struct mv_engine {
     int device = 0, rank = 0, nranks = 1;
     ncclComm_t comm = nullptr;
+    mv_lb_session *lb = nullptr;
     hipStream_t stream = nullptr;
 
     // graph (device)
@@ -1393,6 +1430,33 @@ mv_engine *mv_engine_create(int device, int rank, int nranks,
         std::memcpy(&id, comm_id, sizeof(id));
         NCCL_CHECK(ncclCommInitRank(&e->comm, nranks, id, rank));
     }
+    return e;
+}
+
+mv_lb_session *mv_lb_create(int nranks) { return new mv_lb_session(nranks); }
+
+void mv_lb_destroy(mv_lb_session *s) { delete s; }
+
+mv_engine *mv_engine_create_lb(int device, int rank, int nranks,
+                               mv_lb_session *s) {
+    int ndev = 0;
+    if (hipGetDeviceCount(&ndev) != hipSuccess || ndev <= device) {
+        std::fprintf(stderr,
+                     "mv_engine_create_lb: no HIP device %d (found %d)\n",
+                     device, ndev);
+        return nullptr;
+    }
+    if (!s || s->nranks != nranks) {
+        std::fprintf(stderr, "mv_engine_create_lb: bad session\n");
+        return nullptr;
+    }
+    auto *e = new mv_engine;
+    e->device = device;
+    e->rank = rank;
+    e->nranks = nranks;
+    e->lb = s;
+    HIP_CHECK(hipSetDevice(device));
+    HIP_CHECK(hipStreamCreate(&e->stream));
     return e;
 }
 
@@ -1565,10 +1629,37 @@ void mv_engine_set_trace(mv_engine *e, int64_t *target_trace, double *mod_trace,
 
 void mv_engine_get_stats(const mv_engine *e, mv_stats *out) { *out = e->stats; }
 
-// helper: alltoallv over RCCL grouped send/recv; offsets in elements
+// helper: alltoallv; offsets in elements. RCCL grouped send/recv on the
+// product path, device-to-device copies on the loopback harness.
 static void rccl_alltoallv(mv_engine *e, const void *send, const i64 *soff,
                            void *recv, const i64 *roff, size_t elem_bytes,
                            ncclDataType_t ty, size_t ty_bytes) {
+    if (e->lb) {
+        mv_lb_session *s = e->lb;
+        // own kernels feeding `send` must be complete before peers copy
+        HIP_CHECK(hipStreamSynchronize(e->stream));
+        s->posts[e->rank].send = send;
+        s->posts[e->rank].soff = soff;
+        s->posts[e->rank].eb = elem_bytes;
+        s->barrier();
+        for (int r = 0; r < e->nranks; r++) {
+            if (r == e->rank) continue;
+            const auto &ps = s->posts[r];
+            const i64 cnt = ps.soff[e->rank + 1] - ps.soff[e->rank];
+            if (cnt != roff[r + 1] - roff[r]) {
+                std::fprintf(stderr, "loopback alltoallv count mismatch\n");
+                std::abort();
+            }
+            if (cnt > 0)
+                HIP_CHECK(hipMemcpyAsync(
+                    (char *)recv + roff[r] * elem_bytes,
+                    (const char *)ps.send + ps.soff[e->rank] * elem_bytes,
+                    cnt * elem_bytes, hipMemcpyDeviceToDevice, e->stream));
+        }
+        HIP_CHECK(hipStreamSynchronize(e->stream));
+        s->barrier(); // senders may reuse their buffers after this
+        return;
+    }
     NCCL_CHECK(ncclGroupStart());
     for (int r = 0; r < e->nranks; r++) {
         if (r == e->rank) continue;
@@ -1589,17 +1680,49 @@ static void rccl_alltoallv(mv_engine *e, const void *send, const i64 *soff,
 // exchange per-peer counts: allgather of my nranks counts
 static void exchange_counts(mv_engine *e, const std::vector<i64> &mine,
                             std::vector<i64> &matrix /*nranks*nranks*/) {
+    matrix.resize((size_t)e->nranks * e->nranks);
+    if (e->lb) {
+        mv_lb_session *s = e->lb;
+        s->posts[e->rank].cnts = mine;
+        s->barrier();
+        for (int r = 0; r < e->nranks; r++)
+            std::copy(s->posts[r].cnts.begin(), s->posts[r].cnts.end(),
+                      matrix.begin() + (size_t)r * e->nranks);
+        s->barrier();
+        return;
+    }
     i64 *d_all = nullptr;
     HIP_CHECK(hipMalloc(&d_all, 8 * e->nranks * e->nranks));
     HIP_CHECK(hipMemcpyAsync(d_all + (i64)e->rank * e->nranks, mine.data(),
                              8 * e->nranks, hipMemcpyHostToDevice, e->stream));
     NCCL_CHECK(ncclAllGather(d_all + (i64)e->rank * e->nranks, d_all,
                              e->nranks, ncclInt64, e->comm, e->stream));
-    matrix.resize((size_t)e->nranks * e->nranks);
     HIP_CHECK(hipMemcpyAsync(matrix.data(), d_all, 8 * e->nranks * e->nranks,
                              hipMemcpyDeviceToHost, e->stream));
     HIP_CHECK(hipStreamSynchronize(e->stream));
     HIP_CHECK(hipFree(d_all));
+}
+
+// allreduce-sum of n (<= 2) host doubles in place (dspl.hpp:126, :441)
+static void comm_allreduce_host(mv_engine *e, double *vals, int n) {
+    if (e->lb) {
+        mv_lb_session *s = e->lb;
+        s->posts[e->rank].red = vals;
+        s->barrier();
+        double acc[2] = {0.0, 0.0};
+        for (int r = 0; r < e->nranks; r++) // rank order: deterministic
+            for (int k = 0; k < n; k++) acc[k] += s->posts[r].red[k];
+        s->barrier(); // all ranks have read every post
+        for (int k = 0; k < n; k++) vals[k] = acc[k];
+        return;
+    }
+    HIP_CHECK(hipMemcpyAsync(e->d_red, vals, 8 * n, hipMemcpyHostToDevice,
+                             e->stream));
+    NCCL_CHECK(ncclAllReduce(e->d_red, e->d_red, n, ncclDouble, ncclSum,
+                             e->comm, e->stream));
+    HIP_CHECK(hipMemcpyAsync(vals, e->d_red, 8 * n, hipMemcpyDeviceToHost,
+                             e->stream));
+    HIP_CHECK(hipStreamSynchronize(e->stream));
 }
 
 // Build the SELL image, internal order, spill extents and the high-degree
@@ -1879,13 +2002,7 @@ double mv_engine_run(mv_engine *e, double lower, double thresh,
     double localW = 0.0;
     for (int b = 0; b < nblocks; b++) localW += partials[2 * b];
     double totalW = localW;
-    if (p > 1) { // allreduce (dspl.hpp:126)
-        HIP_CHECK(hipMemcpyAsync(e->d_red, &localW, 8, hipMemcpyHostToDevice, st));
-        NCCL_CHECK(ncclAllReduce(e->d_red, e->d_red, 1, ncclDouble, ncclSum,
-                                 e->comm, st));
-        HIP_CHECK(hipMemcpyAsync(&totalW, e->d_red, 8, hipMemcpyDeviceToHost, st));
-        HIP_CHECK(hipStreamSynchronize(st));
-    }
+    if (p > 1) comm_allreduce_host(e, &totalW, 1); // allreduce (dspl.hpp:126)
     const double constant = 1.0 / totalW; // dspl.hpp:129
     PHASE("k2-done");
     if (p == 1) // slot mode: the i64 arrays alias u32 slot arrays
@@ -2186,15 +2303,7 @@ double mv_engine_run(mv_engine *e, double lower, double thresh,
             la += partials[2 * b + 1];
         }
         double red[2] = {le, la};
-        if (p > 1) {
-            HIP_CHECK(hipMemcpyAsync(e->d_red, red, 16, hipMemcpyHostToDevice,
-                                     st));
-            NCCL_CHECK(ncclAllReduce(e->d_red, e->d_red, 2, ncclDouble,
-                                     ncclSum, e->comm, st));
-            HIP_CHECK(hipMemcpyAsync(red, e->d_red, 16, hipMemcpyDeviceToHost,
-                                     st));
-            HIP_CHECK(hipStreamSynchronize(st));
-        }
+        if (p > 1) comm_allreduce_host(e, red, 2);
         currMod = std::fabs(red[0] * constant - red[1] * constant * constant);
         if (getenv("MV_MOD_DEBUG"))
             std::fprintf(stderr, "[mod] iter=%d le=%.17g la=%.17g\n",
