@@ -441,7 +441,7 @@ __global__ void k_depermute(i64 lnv, const unsigned *__restrict__ sigma_inv,
 // counter[0], dspl.hpp:312-318).
 template <int SLOTS, bool UNIT>
 __global__ __launch_bounds__(256) void k4_sweep(
-    i64 s_begin, i64 lnv, i64 base, i64 bound,
+    i64 s_begin, i64 s_end, i64 lnv, i64 base, i64 bound,
     const unsigned *__restrict__ perm,
     const unsigned *__restrict__ deg_int, const i64 *__restrict__ chunk_off,
     const int *__restrict__ sell_tidx, const double *__restrict__ sell_w,
@@ -463,7 +463,7 @@ __global__ __launch_bounds__(256) void k4_sweep(
     i64 *myspill_k = spill_keys + spill_off[gthread];
     double *myspill_a = spill_acc + spill_off[gthread];
 
-    for (i64 s = s_begin + gthread; s < lnv; s += stride) {
+    for (i64 s = s_begin + gthread; s < s_end; s += stride) {
         const i64 i = perm[s];          // internal vertex index
         const int deg = (int)deg_int[i];
         const i64 ebase = chunk_off[s >> 6] + (s & 63);
@@ -787,7 +787,7 @@ __global__ __launch_bounds__(256) void k4_sweep_hi(
 // (ay-0.0) round exactly like the reference's expressions.
 template <bool UNIT>
 __global__ __launch_bounds__(256) void k4_sweep_iter1(
-    i64 s_begin, i64 lnv, i64 base, i64 bound,
+    i64 s_begin, i64 s_end, i64 lnv, i64 base, i64 bound,
     const unsigned *__restrict__ perm,
     const unsigned *__restrict__ deg_int, const i64 *__restrict__ chunk_off,
     const int *__restrict__ sell_tidx, const double *__restrict__ sell_w,
@@ -799,7 +799,7 @@ __global__ __launch_bounds__(256) void k4_sweep_iter1(
     double *__restrict__ clusterWeight) {
     const i64 gthread = blockIdx.x * (i64)blockDim.x + threadIdx.x;
     const i64 stride = (i64)gridDim.x * blockDim.x;
-    for (i64 s = s_begin + gthread; s < lnv; s += stride) {
+    for (i64 s = s_begin + gthread; s < s_end; s += stride) {
         const i64 i = perm[s];
         const int deg = (int)deg_int[i];
         const i64 ebase = chunk_off[s >> 6] + (s & 63);
@@ -1295,8 +1295,14 @@ struct mv_lb_session {
 struct mv_engine {
     int device = 0, rank = 0, nranks = 1;
     ncclComm_t comm = nullptr;
+    ncclComm_t comm2 = nullptr; // halo #1a's communicator (overlap mode)
     mv_lb_session *lb = nullptr;
     hipStream_t stream = nullptr;
+    hipStream_t stream2 = nullptr; // halo #1a rides here, overlapped
+    hipEvent_t ev_p1 = nullptr;    // sweep part 1 (exports) done [stream]
+    hipEvent_t ev_halo = nullptr;  // #1a recv complete [stream2]
+    int overlap = 0;               // this run overlaps #1a (p>1, !skewed)
+    i64 nexp = 0;                  // SELL positions [0,nexp) = exported
 
     // graph (device)
     i64 nv = 0, lnv = 0, lne = 0, base = 0, bound = 0;
@@ -1334,7 +1340,8 @@ struct mv_engine {
     // ghosts / halo
     i64 *d_ghosts = nullptr;     // sorted unique remote tails
     i64 nghost = 0;
-    i64 *d_ghost_comm = nullptr; // per-iteration communities of ghosts
+    i64 *d_gc[2] = {nullptr, nullptr}; // ghost communities (double-
+                                       // buffered for #1a overlap)
     i64 *d_svdata = nullptr;     // vertices peers want from me (global ids)
     unsigned *d_svdata_int = nullptr;
     i64 ssz = 0;
@@ -1429,7 +1436,15 @@ mv_engine *mv_engine_create(int device, int rank, int nranks,
         ncclUniqueId id;
         std::memcpy(&id, comm_id, sizeof(id));
         NCCL_CHECK(ncclCommInitRank(&e->comm, nranks, id, rank));
+        // halo #1a rides its own communicator so RCCL can run it
+        // concurrently with the epilogue's collectives (MV_NO_OVERLAP must
+        // be set on ALL ranks or none — ncclCommSplit is collective)
+        if (!getenv("MV_NO_OVERLAP"))
+            NCCL_CHECK(ncclCommSplit(e->comm, 0, rank, &e->comm2, nullptr));
     }
+    HIP_CHECK(hipStreamCreate(&e->stream2));
+    HIP_CHECK(hipEventCreate(&e->ev_p1));
+    HIP_CHECK(hipEventCreate(&e->ev_halo));
     return e;
 }
 
@@ -1457,6 +1472,9 @@ mv_engine *mv_engine_create_lb(int device, int rank, int nranks,
     e->lb = s;
     HIP_CHECK(hipSetDevice(device));
     HIP_CHECK(hipStreamCreate(&e->stream));
+    HIP_CHECK(hipStreamCreate(&e->stream2));
+    HIP_CHECK(hipEventCreate(&e->ev_p1));
+    HIP_CHECK(hipEventCreate(&e->ev_halo));
     return e;
 }
 
@@ -1464,8 +1482,12 @@ void mv_engine_destroy(mv_engine *e) {
     if (!e) return;
     (void)hipSetDevice(e->device); // teardown: best effort
     for (auto ev : e->ev_pool) (void)hipEventDestroy(ev);
+    if (e->ev_p1) (void)hipEventDestroy(e->ev_p1);
+    if (e->ev_halo) (void)hipEventDestroy(e->ev_halo);
+    if (e->comm2) ncclCommDestroy(e->comm2);
     if (e->comm) ncclCommDestroy(e->comm);
     free_graph_state(e);
+    if (e->stream2) (void)hipStreamDestroy(e->stream2);
     if (e->stream) (void)hipStreamDestroy(e->stream);
     delete e;
 }
@@ -1485,7 +1507,8 @@ static void free_graph_state(mv_engine *e) {
                      (void **)&e->d_cupd, (void **)&e->d_partials,
                      (void **)&e->d_red, (void **)&e->d_count,
                      (void **)&e->d_bounds, (void **)&e->d_ghosts,
-                     (void **)&e->d_ghost_comm, (void **)&e->d_svdata,
+                     (void **)&e->d_gc[0], (void **)&e->d_gc[1],
+                     (void **)&e->d_svdata,
                      (void **)&e->d_svdata_int, (void **)&e->d_scdata,
                      (void **)&e->d_cand, (void **)&e->d_cand_sorted,
                      (void **)&e->d_rc_ids, (void **)&e->d_rc_info,
@@ -1588,8 +1611,7 @@ int mv_engine_load_graph(mv_engine *e, const mv_graph *g) {
     HIP_CHECK(hipMalloc(&e->d_cupd, sizeof(Cinfo) * lnv));
     // never-null dummies for pointers that stay unused at nranks==1 but
     // may still be address-computed by if-converted loads
-    if (!e->d_ghost_comm)
-        HIP_CHECK(hipMalloc(&e->d_ghost_comm, 16));
+    if (!e->d_gc[0]) HIP_CHECK(hipMalloc(&e->d_gc[0], 16));
     if (!e->d_rc_ids) HIP_CHECK(hipMalloc(&e->d_rc_ids, 16));
     if (!e->d_rc_info) HIP_CHECK(hipMalloc(&e->d_rc_info, sizeof(Info16)));
     if (!e->d_rcu) HIP_CHECK(hipMalloc(&e->d_rcu, sizeof(Info16)));
@@ -1633,11 +1655,12 @@ void mv_engine_get_stats(const mv_engine *e, mv_stats *out) { *out = e->stats; }
 // product path, device-to-device copies on the loopback harness.
 static void rccl_alltoallv(mv_engine *e, const void *send, const i64 *soff,
                            void *recv, const i64 *roff, size_t elem_bytes,
-                           ncclDataType_t ty, size_t ty_bytes) {
+                           ncclDataType_t ty, size_t ty_bytes,
+                           ncclComm_t comm, hipStream_t stream) {
     if (e->lb) {
         mv_lb_session *s = e->lb;
         // own kernels feeding `send` must be complete before peers copy
-        HIP_CHECK(hipStreamSynchronize(e->stream));
+        HIP_CHECK(hipStreamSynchronize(stream));
         s->posts[e->rank].send = send;
         s->posts[e->rank].soff = soff;
         s->posts[e->rank].eb = elem_bytes;
@@ -1654,9 +1677,9 @@ static void rccl_alltoallv(mv_engine *e, const void *send, const i64 *soff,
                 HIP_CHECK(hipMemcpyAsync(
                     (char *)recv + roff[r] * elem_bytes,
                     (const char *)ps.send + ps.soff[e->rank] * elem_bytes,
-                    cnt * elem_bytes, hipMemcpyDeviceToDevice, e->stream));
+                    cnt * elem_bytes, hipMemcpyDeviceToDevice, stream));
         }
-        HIP_CHECK(hipStreamSynchronize(e->stream));
+        HIP_CHECK(hipStreamSynchronize(stream));
         s->barrier(); // senders may reuse their buffers after this
         return;
     }
@@ -1667,12 +1690,12 @@ static void rccl_alltoallv(mv_engine *e, const void *send, const i64 *soff,
         const i64 rcnt = roff[r + 1] - roff[r];
         if (scnt > 0)
             NCCL_CHECK(ncclSend((const char *)send + soff[r] * elem_bytes,
-                                scnt * elem_bytes / ty_bytes, ty, r, e->comm,
-                                e->stream));
+                                scnt * elem_bytes / ty_bytes, ty, r, comm,
+                                stream));
         if (rcnt > 0)
             NCCL_CHECK(ncclRecv((char *)recv + roff[r] * elem_bytes,
-                                rcnt * elem_bytes / ty_bytes, ty, r, e->comm,
-                                e->stream));
+                                rcnt * elem_bytes / ty_bytes, ty, r, comm,
+                                stream));
     }
     NCCL_CHECK(ncclGroupEnd());
 }
@@ -1792,6 +1815,40 @@ static void build_sell(mv_engine *e) {
         HIP_CHECK(hipFree(d_tmp));
         HIP_CHECK(hipFree(d_degs));
     }
+
+    // Exports-first SELL order (halo overlap): positions [0, nexp) hold
+    // the vertices some peer wants (the svdata set), so the sweep can run
+    // them as a first launch and halo #1a for the NEXT iteration — which
+    // only needs their targets — can overlap the interior sweep + the
+    // whole epilogue on stream2/comm2 (north_star: "#1a overlapped with
+    // the local sweep on a second HIP stream"). Stable partition: locality
+    // inside each group is preserved. Pure layout, results identical.
+    e->nexp = 0;
+    e->overlap = 0;
+    if (e->nranks > 1 && !e->skewed && e->ssz > 0 && (e->comm2 || e->lb) &&
+        !getenv("MV_NO_OVERLAP")) {
+        std::vector<unsigned> perm_h(lnv);
+        HIP_CHECK(hipMemcpyAsync(perm_h.data(), e->d_perm, 4 * lnv,
+                                 hipMemcpyDeviceToHost, st));
+        std::vector<unsigned> sv(e->ssz);
+        HIP_CHECK(hipMemcpyAsync(sv.data(), e->d_svdata_int, 4 * e->ssz,
+                                 hipMemcpyDeviceToHost, st));
+        HIP_CHECK(hipStreamSynchronize(st));
+        std::vector<char> mark(lnv, 0);
+        for (i64 k = 0; k < e->ssz; k++) mark[sv[k]] = 1;
+        std::vector<unsigned> nperm(lnv);
+        i64 w = 0;
+        for (i64 s = 0; s < lnv; s++)
+            if (mark[perm_h[s]]) nperm[w++] = perm_h[s];
+        e->nexp = w;
+        for (i64 s = 0; s < lnv; s++)
+            if (!mark[perm_h[s]]) nperm[w++] = perm_h[s];
+        HIP_CHECK(hipMemcpyAsync(e->d_perm, nperm.data(), 4 * lnv,
+                                 hipMemcpyHostToDevice, st));
+        HIP_CHECK(hipStreamSynchronize(st));
+        e->overlap = 1;
+    }
+
     {
         i64 *d_sizes = nullptr;
         HIP_CHECK(hipMalloc(&d_sizes, 8 * std::max<i64>(e->nchunks, 1)));
@@ -1963,14 +2020,17 @@ double mv_engine_run(mv_engine *e, double lower, double thresh,
             // role swap (dspl.hpp:1255-1257): my ghost list goes OUT, the
             // peers' lists land in svdata
             rccl_alltoallv(e, e->d_ghosts, e->recv_off.data(), e->d_svdata,
-                           e->send_off.data(), 8, ncclInt64, 8);
+                           e->send_off.data(), 8, ncclInt64, 8, e->comm,
+                           e->stream);
             k_to_internal<<<grid_for(std::max<i64>(e->ssz, 1)), 256, 0, st>>>(
                 e->ssz, e->d_svdata, e->base, e->d_sigma_inv, e->d_svdata_int);
             HIP_CHECK(hipStreamSynchronize(st));
 
-            if (e->d_ghost_comm) HIP_CHECK(hipFree(e->d_ghost_comm));
-            HIP_CHECK(hipMalloc(&e->d_ghost_comm,
-                                8 * std::max<i64>(e->nghost, 1)));
+            for (int b = 0; b < 2; b++) {
+                if (e->d_gc[b]) HIP_CHECK(hipFree(e->d_gc[b]));
+                HIP_CHECK(hipMalloc(&e->d_gc[b],
+                                    8 * std::max<i64>(e->nghost, 1)));
+            }
             if (e->d_scdata) HIP_CHECK(hipFree(e->d_scdata));
             HIP_CHECK(hipMalloc(&e->d_scdata, 8 * std::max<i64>(e->ssz, 1)));
         } else {
@@ -2021,20 +2081,46 @@ double mv_engine_run(mv_engine *e, double lower, double thresh,
     i64 *d_curr = e->d_curr, *d_past = e->d_past, *d_target = e->d_target;
     double prevMod = lower, currMod = -1.0;
     int numIters = 0;
+    int gcur = 0; // which d_gc buffer THIS iteration's ghost comms live in
     std::vector<i64> rc_bounds(p + 1, 0), req_off(p + 1, 0);
+
+    // Overlap mode: gather + send the NEXT iteration's ghost communities
+    // (halo #1a, dspl.hpp:583-647) on stream2/comm2 as soon as the export
+    // positions' targets are written (sweep part 1), hiding the exchange
+    // behind the interior sweep and the epilogue. Joined via ev_halo at
+    // the next iteration's top.
+    auto issue_halo1a = [&](const i64 *commArr, i64 *dst) {
+        HIP_CHECK(hipEventRecord(e->ev_p1, st));
+        HIP_CHECK(hipStreamWaitEvent(e->stream2, e->ev_p1, 0));
+        k8_gather_comms<<<grid_for(std::max<i64>(e->ssz, 1)), 256, 0,
+                          e->stream2>>>(e->ssz, e->d_svdata_int, commArr,
+                                        e->d_scdata);
+        rccl_alltoallv(e, e->d_scdata, e->send_off.data(), dst,
+                       e->recv_off.data(), 8, ncclInt64, 8,
+                       e->comm2 ? e->comm2 : e->comm, e->stream2);
+        HIP_CHECK(hipEventRecord(e->ev_halo, e->stream2));
+    };
+    if (p > 1 && e->overlap) issue_halo1a(e->d_curr, e->d_gc[0]);
 
     std::vector<hipEvent_t> sweep_ev;
     for (;;) {
         numIters++;
 
         i64 nrc = 0;
+        i64 *gc = e->d_gc[gcur];
         if (p > 1) {
             const auto t_h0 = std::chrono::steady_clock::now();
             // ---- halo #1a: ghost communities (dspl.hpp:583-647) ----
-            k8_gather_comms<<<grid_for(std::max<i64>(e->ssz, 1)), 256, 0, st>>>(
-                e->ssz, e->d_svdata_int, d_curr, e->d_scdata);
-            rccl_alltoallv(e, e->d_scdata, e->send_off.data(), e->d_ghost_comm,
-                           e->recv_off.data(), 8, ncclInt64, 8);
+            if (e->overlap) {
+                HIP_CHECK(hipStreamWaitEvent(st, e->ev_halo, 0));
+            } else {
+                k8_gather_comms<<<grid_for(std::max<i64>(e->ssz, 1)), 256, 0,
+                                  st>>>(e->ssz, e->d_svdata_int, d_curr,
+                                        e->d_scdata);
+                rccl_alltoallv(e, e->d_scdata, e->send_off.data(), gc,
+                               e->recv_off.data(), 8, ncclInt64, 8, e->comm,
+                               e->stream);
+            }
 
             // ---- needed remote communities (dspl.hpp:670-700) ----
             const i64 cand_max = e->nghost + lnv;
@@ -2065,7 +2151,7 @@ double mv_engine_run(mv_engine *e, double lower, double thresh,
             }
             HIP_CHECK(hipMemsetAsync(e->d_count, 0, 8, st));
             k_filter_remote<<<grid_for(std::max<i64>(e->nghost, 1)), 256, 0,
-                              st>>>(e->nghost, e->d_ghost_comm, /*shift*/ 0,
+                              st>>>(e->nghost, gc, /*shift*/ 0,
                                     e->base, e->bound, e->d_cand, e->d_count);
             k_filter_remote<<<grid_for(lnv), 256, 0, st>>>(
                 lnv, d_curr, /*shift*/ 32, e->base, e->bound, e->d_cand,
@@ -2086,7 +2172,7 @@ double mv_engine_run(mv_engine *e, double lower, double thresh,
             // ghost communities: labels -> handles for the sweep
             k_labels_to_handles<<<grid_for(std::max<i64>(e->nghost, 1)), 256,
                                   0, st>>>(e->nghost, e->base, e->bound,
-                                           e->d_sigma_inv, e->d_ghost_comm);
+                                           e->d_sigma_inv, gc);
             HIP_CHECK(hipStreamSynchronize(st));
 
             // ---- halo #1b/#1c/#1d: request (size,degree) of those
@@ -2114,12 +2200,14 @@ double mv_engine_run(mv_engine *e, double lower, double thresh,
                     hipMalloc(&e->d_req_info, sizeof(Info16) * e->req_cap));
             }
             rccl_alltoallv(e, e->d_rc_ids, rc_bounds.data(), e->d_req_ids,
-                           req_off.data(), 8, ncclInt64, 8);
+                           req_off.data(), 8, ncclInt64, 8, e->comm,
+                           e->stream);
             k9_reply_info<<<grid_for(std::max<i64>(nreq, 1)), 256, 0, st>>>(
                 nreq, e->d_req_ids, e->base, e->d_sigma_inv, e->d_cinfo,
                 e->d_req_info);
             rccl_alltoallv(e, e->d_req_info, req_off.data(), e->d_rc_info,
-                           rc_bounds.data(), sizeof(Info16), ncclChar, 1);
+                           rc_bounds.data(), sizeof(Info16), ncclChar, 1,
+                           e->comm, e->stream);
             HIP_CHECK(hipMemsetAsync(e->d_rcu, 0,
                                      sizeof(Info16) * std::max<i64>(nrc, 1),
                                      st));
@@ -2163,12 +2251,13 @@ double mv_engine_run(mv_engine *e, double lower, double thresh,
             return s ? atoi(s) : 2;
         }();
         const int slots = (numIters <= first_iters) ? slots_first : slots_rest;
-        auto launch_sweep = [&](auto slots_tag, auto unit_tag) {
+        auto launch_sweep = [&](auto slots_tag, auto unit_tag, i64 s0,
+                                i64 s1) {
             constexpr int S = decltype(slots_tag)::value;
             if (p == 1) { // u32 slots: half-size gathers, 12 B/lane LDS
                 k4_sweep_p1<S, decltype(unit_tag)::value>
                     <<<e->sweep_grid, 256, S * 256 * 12, st>>>(
-                        e->nhi, lnv, e->d_perm, e->d_deg, e->d_chunk_off,
+                        s0, lnv, e->d_perm, e->d_deg, e->d_chunk_off,
                         e->d_sell_tidx, e->d_sell_w, (const unsigned *)d_curr,
                         e->d_vdeg, e->d_sigma, e->d_cinfo, e->d_cupd,
                         constant, (unsigned *)d_target, e->d_cw,
@@ -2177,38 +2266,38 @@ double mv_engine_run(mv_engine *e, double lower, double thresh,
                 return;
             }
             k4_sweep<S, decltype(unit_tag)::value>
-                <<<e->sweep_grid, 256, S * 256 * 16, st>>>(
-                    e->nhi, lnv, e->base, e->bound, e->d_perm, e->d_deg,
+                <<<grid_for(s1 - s0, 256, 2048), 256, S * 256 * 16, st>>>(
+                    s0, s1, lnv, e->base, e->bound, e->d_perm, e->d_deg,
                     e->d_chunk_off, e->d_sell_tidx, e->d_sell_w, d_curr,
-                    e->d_ghost_comm, e->d_vdeg, e->d_sigma, e->d_cinfo,
+                    gc, e->d_vdeg, e->d_sigma, e->d_cinfo,
                     e->d_cupd, e->d_rc_ids, nrc, e->d_rc_info, e->d_rcu,
                     constant, d_target, e->d_cw, e->d_spill_k, e->d_spill_a,
                     e->d_spill_off,
                     getenv("MV_DBG_SKIP") ? atoi(getenv("MV_DBG_SKIP")) : 0);
         };
-        auto dispatch_slots = [&](auto unit_tag) {
+        auto dispatch_slots = [&](auto unit_tag, i64 s0, i64 s1) {
             switch (slots) {
-            case 4: launch_sweep(std::integral_constant<int, 4>{}, unit_tag); break;
-            case 12: launch_sweep(std::integral_constant<int, 12>{}, unit_tag); break;
-            case 16: launch_sweep(std::integral_constant<int, 16>{}, unit_tag); break;
-            case 24: launch_sweep(std::integral_constant<int, 24>{}, unit_tag); break;
-            default: launch_sweep(std::integral_constant<int, 8>{}, unit_tag); break;
+            case 4: launch_sweep(std::integral_constant<int, 4>{}, unit_tag, s0, s1); break;
+            case 12: launch_sweep(std::integral_constant<int, 12>{}, unit_tag, s0, s1); break;
+            case 16: launch_sweep(std::integral_constant<int, 16>{}, unit_tag, s0, s1); break;
+            case 24: launch_sweep(std::integral_constant<int, 24>{}, unit_tag, s0, s1); break;
+            default: launch_sweep(std::integral_constant<int, 8>{}, unit_tag, s0, s1); break;
             }
         };
-        auto launch_iter1 = [&](auto unit_tag) {
+        auto launch_iter1 = [&](auto unit_tag, i64 s0, i64 s1) {
             if (p == 1) {
                 k4_sweep_iter1_p1<decltype(unit_tag)::value>
                     <<<e->sweep_grid, 256, 0, st>>>(
-                        e->nhi, lnv, e->d_perm, e->d_deg, e->d_chunk_off,
+                        s0, lnv, e->d_perm, e->d_deg, e->d_chunk_off,
                         e->d_sell_tidx, e->d_sell_w, e->d_vdeg, e->d_sigma,
                         e->d_cupd, constant, (unsigned *)d_target, e->d_cw);
                 return;
             }
             k4_sweep_iter1<decltype(unit_tag)::value>
-                <<<e->sweep_grid, 256, 0, st>>>(
-                    e->nhi, lnv, e->base, e->bound, e->d_perm, e->d_deg,
+                <<<grid_for(s1 - s0, 256, 2048), 256, 0, st>>>(
+                    s0, s1, lnv, e->base, e->bound, e->d_perm, e->d_deg,
                     e->d_chunk_off, e->d_sell_tidx, e->d_sell_w, d_curr,
-                    e->d_ghost_comm, e->d_vdeg, e->d_cupd, e->d_rc_ids, nrc,
+                    gc, e->d_vdeg, e->d_cupd, e->d_rc_ids, nrc,
                     e->d_rc_info, e->d_rcu, constant, d_target, e->d_cw);
         };
         if (e->nhi > 0) { // wave-per-vertex hubs (unit weights only)
@@ -2227,21 +2316,33 @@ double mv_engine_run(mv_engine *e, double lower, double thresh,
                                 st>>>(
                 e->nhi, lnv, e->base, e->bound, e->d_perm, e->d_deg,
                 e->d_sigma, e->d_sigma_inv, e->d_xadj, e->d_tails, e->d_ew,
-                e->d_ghosts, e->nghost, d_curr, e->d_ghost_comm, e->d_vdeg,
+                e->d_ghosts, e->nghost, d_curr, gc, e->d_vdeg,
                 e->d_cinfo, e->d_cupd, e->d_rc_ids, nrc, e->d_rc_info,
                 e->d_rcu, constant, d_target, e->d_cw, e->d_hash_off,
                 e->d_hkeys, e->d_hacc);
         }
         static const bool no_iter1 = getenv("MV_NO_ITER1") != nullptr;
-        if (numIters == 1 && e->rows_sorted && !no_iter1) {
-            if (e->unit_weights)
-                launch_iter1(std::integral_constant<bool, true>{});
-            else
-                launch_iter1(std::integral_constant<bool, false>{});
-        } else if (e->unit_weights) {
-            dispatch_slots(std::integral_constant<bool, true>{});
+        auto sweep_range = [&](i64 s0, i64 s1) {
+            if (s1 <= s0) return;
+            if (numIters == 1 && e->rows_sorted && !no_iter1) {
+                if (e->unit_weights)
+                    launch_iter1(std::integral_constant<bool, true>{}, s0, s1);
+                else
+                    launch_iter1(std::integral_constant<bool, false>{}, s0, s1);
+            } else if (e->unit_weights) {
+                dispatch_slots(std::integral_constant<bool, true>{}, s0, s1);
+            } else {
+                dispatch_slots(std::integral_constant<bool, false>{}, s0, s1);
+            }
+        };
+        if (p > 1 && e->overlap) {
+            // exports first, then launch the NEXT iteration's #1a off their
+            // targets while the interior positions sweep
+            sweep_range(0, e->nexp);
+            issue_halo1a(d_target, e->d_gc[gcur ^ 1]);
+            sweep_range(e->nexp, lnv);
         } else {
-            dispatch_slots(std::integral_constant<bool, false>{});
+            sweep_range(e->nhi, lnv);
         }
         HIP_CHECK(hipEventRecord(ev1, st));
         PHASE("sweep-done");
@@ -2263,7 +2364,8 @@ double mv_engine_run(mv_engine *e, double lower, double thresh,
         if (p > 1) {
             const auto t_h0 = std::chrono::steady_clock::now();
             rccl_alltoallv(e, e->d_rcu, rc_bounds.data(), e->d_req_info,
-                           req_off.data(), sizeof(Info16), ncclChar, 1);
+                           req_off.data(), sizeof(Info16), ncclChar, 1,
+                           e->comm, e->stream);
             // apply per SENDER segment, in rank order: same-stream launches
             // serialize, and one sender's ids are unique (sorted rc_ids), so
             // cinfo degree bits are run-to-run identical at any nranks —
@@ -2336,8 +2438,13 @@ double mv_engine_run(mv_engine *e, double lower, double thresh,
         d_past = d_curr;
         d_curr = d_target;
         d_target = tmp;
+        if (e->overlap) gcur ^= 1; // next iteration reads the #1a buffer
+                                   // filled during this one
         if (numIters >= 10000) break; // safety net, never hit in practice
     }
+    // overlap mode posted one speculative #1a past the exit; every rank
+    // posted it symmetrically — drain it before returning
+    if (e->overlap) HIP_CHECK(hipStreamSynchronize(e->stream2));
 
     const bool dbg = getenv("MV_SWEEP_DEBUG") != nullptr;
     for (size_t k = 0; k + 1 < sweep_ev.size() + 1; k += 2) {
