@@ -306,17 +306,53 @@ __global__ void k_to_internal(i64 n, const i64 *__restrict__ gids, i64 base,
         out[k] = sigma_inv[gids[k] - base];
 }
 
-// ghost communities arrive as LABELS; convert to handles
+// ghost communities arrive as LABELS; convert to handles (out-of-place:
+// src stays the persistent label image the delta scatter maintains)
 __global__ void k_labels_to_handles(i64 n, i64 base, i64 bound,
                                     const unsigned *__restrict__ sigma_inv,
-                                    i64 *__restrict__ vals) {
+                                    const i64 *__restrict__ src,
+                                    i64 *__restrict__ dst) {
     for (i64 k = blockIdx.x * (i64)blockDim.x + threadIdx.x; k < n;
          k += (i64)gridDim.x * blockDim.x) {
-        const i64 c = vals[k];
-        vals[k] = (c >= base && c < bound)
-                      ? (c << 32) | (i64)sigma_inv[c - base]
-                      : c << 32;
+        const i64 c = src[k];
+        dst[k] = (c >= base && c < bound)
+                     ? (c << 32) | (i64)sigma_inv[c - base]
+                     : c << 32;
     }
+}
+
+// ---- halo #1a delta compaction ----
+// The reference resends the FULL ghost community set every iteration
+// (dspl.hpp:583-647). Here each export k compares against the label LAST
+// SENT and emits a (index-in-segment, label) pair into its peer segment;
+// the allgathered count matrix then lets both ends of every pair decide
+// full vs compact identically (compact record 12 B vs full 8 B/entry).
+// last_sent is updated unconditionally: in full mode the wire carries
+// scdata itself, so "last sent" == current either way.
+__global__ void k_delta_compact(i64 ssz, const i64 *__restrict__ soff,
+                                int nranks, const i64 *__restrict__ scdata,
+                                i64 *__restrict__ last_sent,
+                                unsigned *__restrict__ chg_idx,
+                                i64 *__restrict__ chg_lab,
+                                unsigned long long *__restrict__ cnt) {
+    for (i64 k = blockIdx.x * (i64)blockDim.x + threadIdx.x; k < ssz;
+         k += (i64)gridDim.x * blockDim.x) {
+        const i64 c = scdata[k];
+        if (last_sent[k] == c) continue;
+        last_sent[k] = c;
+        const int r = (int)dev_lower_bound(soff, nranks + 1, k + 1) - 1;
+        const unsigned long long pos = atomicAdd(&cnt[r], 1ull);
+        chg_idx[soff[r] + pos] = (unsigned)(k - soff[r]);
+        chg_lab[soff[r] + pos] = c;
+    }
+}
+
+__global__ void k_scatter_deltas(i64 n, const unsigned *__restrict__ idx,
+                                 const i64 *__restrict__ lab, i64 seg_base,
+                                 i64 *__restrict__ ghost_labels) {
+    for (i64 k = blockIdx.x * (i64)blockDim.x + threadIdx.x; k < n;
+         k += (i64)gridDim.x * blockDim.x)
+        ghost_labels[seg_base + idx[k]] = lab[k];
 }
 
 // ---- K8: scdata gather (dspl.hpp:559-571) ----
@@ -1272,6 +1308,8 @@ struct mv_lb_session {
     uint64_t gen = 0;
     struct Post {
         const void *send = nullptr;
+        const void *send2 = nullptr; // delta mode: chg_idx
+        const void *send3 = nullptr; // delta mode: chg_lab
         const int64_t *soff = nullptr;
         size_t eb = 0;
         const double *red = nullptr;
@@ -1347,6 +1385,20 @@ struct mv_engine {
     i64 ssz = 0;
     std::vector<i64> send_off, recv_off; // per-peer offsets into svdata / ghosts
     i64 *d_scdata = nullptr;             // packed comms to export
+
+    // halo #1a delta compaction (send side tracks the last label sent per
+    // export; receive side keeps the persistent ghost label image)
+    int use_delta = 0;
+    i64 *d_last_sent = nullptr;            // ssz, init -1 (all changed)
+    unsigned *d_chg_idx = nullptr;         // ssz, segment-based layout
+    i64 *d_chg_lab = nullptr;              // ssz
+    unsigned long long *d_chg_cnt = nullptr; // nranks counters
+    i64 *d_cntmat = nullptr;               // nranks*nranks allgather image
+    i64 *h_cntmat = nullptr;               // pinned mirror
+    i64 *d_soff = nullptr;                 // send_off on device (nranks+1)
+    i64 *d_ghost_labels = nullptr;         // nghost persistent labels
+    unsigned *d_rchg_idx = nullptr;        // nghost recv compact indices
+    i64 *d_rchg_lab = nullptr;             // nghost recv compact labels
 
     // remote community info (per iteration)
     i64 rc_cap = 0;
@@ -1517,6 +1569,11 @@ static void free_graph_state(mv_engine *e) {
                      (void **)&e->d_spill_k, (void **)&e->d_spill_a,
                      (void **)&e->d_spill_off, (void **)&e->d_hash_off,
                      (void **)&e->d_hkeys, (void **)&e->d_hacc,
+                     (void **)&e->d_last_sent, (void **)&e->d_chg_idx,
+                     (void **)&e->d_chg_lab, (void **)&e->d_chg_cnt,
+                     (void **)&e->d_cntmat, (void **)&e->d_soff,
+                     (void **)&e->d_ghost_labels, (void **)&e->d_rchg_idx,
+                     (void **)&e->d_rchg_lab,
                      (void **)&e->d_trace_tmp}) {
         if (*p) {
             HIP_CHECK(hipFree(*p));
@@ -1526,6 +1583,10 @@ static void free_graph_state(mv_engine *e) {
     if (e->h_partials) {
         HIP_CHECK(hipHostFree(e->h_partials));
         e->h_partials = nullptr;
+    }
+    if (e->h_cntmat) {
+        HIP_CHECK(hipHostFree(e->h_cntmat));
+        e->h_cntmat = nullptr;
     }
     // a stale trace against a freed d_trace_tmp (or a new lnv) would fault:
     // re-arm via mv_engine_set_trace after every load
@@ -1933,6 +1994,130 @@ static void build_sell(mv_engine *e) {
     HIP_CHECK(hipStreamSynchronize(st));
 }
 
+// Halo #1a (dspl.hpp:583-647) on (comm, stream): gather the exports'
+// communities, exchange them (full resend like the reference, or
+// delta-compacted), scatter into the persistent label image, and emit the
+// HANDLE image for the sweep into gc_dst. In overlap mode this runs on
+// stream2/comm2 concurrently with the interior sweep + epilogue.
+static void run_halo1a(mv_engine *e, const i64 *commArr, i64 *gc_dst,
+                       ncclComm_t comm, hipStream_t st2) {
+    const int p = e->nranks;
+    k8_gather_comms<<<grid_for(std::max<i64>(e->ssz, 1)), 256, 0, st2>>>(
+        e->ssz, e->d_svdata_int, commArr, e->d_scdata);
+    if (!e->use_delta) {
+        rccl_alltoallv(e, e->d_scdata, e->send_off.data(), e->d_ghost_labels,
+                       e->recv_off.data(), 8, ncclInt64, 8, comm, st2);
+    } else {
+        HIP_CHECK(hipMemsetAsync(e->d_chg_cnt, 0, 8 * p, st2));
+        k_delta_compact<<<grid_for(std::max<i64>(e->ssz, 1)), 256, 0, st2>>>(
+            e->ssz, e->d_soff, p, e->d_scdata, e->d_last_sent, e->d_chg_idx,
+            e->d_chg_lab, e->d_chg_cnt);
+        // tiny count matrix; rides the same stream/comm so it overlaps the
+        // interior sweep in overlap mode (the host blocks only on these
+        // small ops — the interior sweep is already issued)
+        if (e->lb) {
+            std::vector<i64> mine(p), m;
+            HIP_CHECK(hipMemcpyAsync(mine.data(), (i64 *)e->d_chg_cnt, 8 * p,
+                                     hipMemcpyDeviceToHost, st2));
+            HIP_CHECK(hipStreamSynchronize(st2));
+            exchange_counts(e, mine, m);
+            std::copy(m.begin(), m.end(), e->h_cntmat);
+        } else {
+            HIP_CHECK(hipMemcpyAsync(e->d_cntmat + (i64)e->rank * p,
+                                     e->d_chg_cnt, 8 * p,
+                                     hipMemcpyDeviceToDevice, st2));
+            NCCL_CHECK(ncclAllGather(e->d_cntmat + (i64)e->rank * p,
+                                     e->d_cntmat, p, ncclInt64, comm, st2));
+            HIP_CHECK(hipMemcpyAsync(e->h_cntmat, e->d_cntmat, 8 * p * p,
+                                     hipMemcpyDeviceToHost, st2));
+            HIP_CHECK(hipStreamSynchronize(st2));
+        }
+        // mixed full/compact payload: both ends of every pair decide from
+        // the same matrix entry, so the plan is symmetric by construction
+        auto full_mode = [](i64 changed, i64 seg) {
+            return changed * 12 >= seg * 8;
+        };
+        if (e->lb) {
+            mv_lb_session *s = e->lb;
+            HIP_CHECK(hipStreamSynchronize(st2));
+            s->posts[e->rank].send = e->d_scdata;
+            s->posts[e->rank].send2 = e->d_chg_idx;
+            s->posts[e->rank].send3 = e->d_chg_lab;
+            s->posts[e->rank].soff = e->send_off.data();
+            s->barrier();
+            for (int r = 0; r < p; r++) {
+                if (r == e->rank) continue;
+                const auto &ps = s->posts[r];
+                const i64 seg = e->recv_off[r + 1] - e->recv_off[r];
+                const i64 chg = e->h_cntmat[(i64)r * p + e->rank];
+                const i64 sb = ps.soff[e->rank];
+                if (full_mode(chg, seg)) {
+                    if (seg > 0)
+                        HIP_CHECK(hipMemcpyAsync(
+                            e->d_ghost_labels + e->recv_off[r],
+                            (const i64 *)ps.send + sb, 8 * seg,
+                            hipMemcpyDeviceToDevice, st2));
+                } else if (chg > 0) {
+                    HIP_CHECK(hipMemcpyAsync(
+                        e->d_rchg_idx + e->recv_off[r],
+                        (const unsigned *)ps.send2 + sb, 4 * chg,
+                        hipMemcpyDeviceToDevice, st2));
+                    HIP_CHECK(hipMemcpyAsync(
+                        e->d_rchg_lab + e->recv_off[r],
+                        (const i64 *)ps.send3 + sb, 8 * chg,
+                        hipMemcpyDeviceToDevice, st2));
+                }
+            }
+            HIP_CHECK(hipStreamSynchronize(st2));
+            s->barrier();
+        } else {
+            NCCL_CHECK(ncclGroupStart());
+            for (int r = 0; r < p; r++) {
+                if (r == e->rank) continue;
+                const i64 sseg = e->send_off[r + 1] - e->send_off[r];
+                const i64 schg = e->h_cntmat[(i64)e->rank * p + r];
+                if (full_mode(schg, sseg)) {
+                    if (sseg > 0)
+                        NCCL_CHECK(ncclSend(e->d_scdata + e->send_off[r],
+                                            sseg, ncclInt64, r, comm, st2));
+                } else if (schg > 0) {
+                    NCCL_CHECK(ncclSend(e->d_chg_idx + e->send_off[r],
+                                        schg * 4, ncclChar, r, comm, st2));
+                    NCCL_CHECK(ncclSend(e->d_chg_lab + e->send_off[r],
+                                        schg, ncclInt64, r, comm, st2));
+                }
+                const i64 rseg = e->recv_off[r + 1] - e->recv_off[r];
+                const i64 rchg = e->h_cntmat[(i64)r * p + e->rank];
+                if (full_mode(rchg, rseg)) {
+                    if (rseg > 0)
+                        NCCL_CHECK(
+                            ncclRecv(e->d_ghost_labels + e->recv_off[r],
+                                     rseg, ncclInt64, r, comm, st2));
+                } else if (rchg > 0) {
+                    NCCL_CHECK(ncclRecv(e->d_rchg_idx + e->recv_off[r],
+                                        rchg * 4, ncclChar, r, comm, st2));
+                    NCCL_CHECK(ncclRecv(e->d_rchg_lab + e->recv_off[r],
+                                        rchg, ncclInt64, r, comm, st2));
+                }
+            }
+            NCCL_CHECK(ncclGroupEnd());
+        }
+        for (int r = 0; r < p; r++) { // scatter compact segments
+            if (r == e->rank) continue;
+            const i64 rseg = e->recv_off[r + 1] - e->recv_off[r];
+            const i64 rchg = e->h_cntmat[(i64)r * p + e->rank];
+            if (!full_mode(rchg, rseg) && rchg > 0)
+                k_scatter_deltas<<<grid_for(rchg), 256, 0, st2>>>(
+                    rchg, e->d_rchg_idx + e->recv_off[r],
+                    e->d_rchg_lab + e->recv_off[r], e->recv_off[r],
+                    e->d_ghost_labels);
+        }
+    }
+    k_labels_to_handles<<<grid_for(std::max<i64>(e->nghost, 1)), 256, 0,
+                          st2>>>(e->nghost, e->base, e->bound,
+                                 e->d_sigma_inv, e->d_ghost_labels, gc_dst);
+}
+
 #define PHASE(tag)                                                            \
     do {                                                                      \
         if (getenv("MV_PHASE_DEBUG")) {                                       \
@@ -2033,6 +2218,37 @@ double mv_engine_run(mv_engine *e, double lower, double thresh,
             }
             if (e->d_scdata) HIP_CHECK(hipFree(e->d_scdata));
             HIP_CHECK(hipMalloc(&e->d_scdata, 8 * std::max<i64>(e->ssz, 1)));
+
+            // delta-compaction state (default on; MV_NO_DELTA reverts to
+            // the reference's full resend for A/B)
+            e->use_delta = !getenv("MV_NO_DELTA");
+            const i64 sszc = std::max<i64>(e->ssz, 1);
+            const i64 ngc = std::max<i64>(e->nghost, 1);
+            for (void **q :
+                 {(void **)&e->d_last_sent, (void **)&e->d_chg_idx,
+                  (void **)&e->d_chg_lab, (void **)&e->d_chg_cnt,
+                  (void **)&e->d_cntmat, (void **)&e->d_soff,
+                  (void **)&e->d_ghost_labels, (void **)&e->d_rchg_idx,
+                  (void **)&e->d_rchg_lab})
+                if (*q) {
+                    HIP_CHECK(hipFree(*q));
+                    *q = nullptr;
+                }
+            HIP_CHECK(hipMalloc(&e->d_last_sent, 8 * sszc));
+            HIP_CHECK(hipMemsetAsync(e->d_last_sent, 0xFF, 8 * sszc, st));
+            HIP_CHECK(hipMalloc(&e->d_chg_idx, 4 * sszc));
+            HIP_CHECK(hipMalloc(&e->d_chg_lab, 8 * sszc));
+            HIP_CHECK(hipMalloc(&e->d_chg_cnt, 8 * p));
+            HIP_CHECK(hipMalloc(&e->d_cntmat, 8 * p * p));
+            HIP_CHECK(hipMalloc(&e->d_soff, 8 * (p + 1)));
+            HIP_CHECK(hipMemcpyAsync(e->d_soff, e->send_off.data(),
+                                     8 * (p + 1), hipMemcpyHostToDevice, st));
+            HIP_CHECK(hipMalloc(&e->d_ghost_labels, 8 * ngc));
+            HIP_CHECK(hipMalloc(&e->d_rchg_idx, 4 * ngc));
+            HIP_CHECK(hipMalloc(&e->d_rchg_lab, 8 * ngc));
+            if (!e->h_cntmat)
+                HIP_CHECK(hipHostMalloc(&e->h_cntmat, 8 * p * p));
+            HIP_CHECK(hipStreamSynchronize(st));
         } else {
             e->nghost = 0;
             e->ssz = 0;
@@ -2084,23 +2300,17 @@ double mv_engine_run(mv_engine *e, double lower, double thresh,
     int gcur = 0; // which d_gc buffer THIS iteration's ghost comms live in
     std::vector<i64> rc_bounds(p + 1, 0), req_off(p + 1, 0);
 
-    // Overlap mode: gather + send the NEXT iteration's ghost communities
-    // (halo #1a, dspl.hpp:583-647) on stream2/comm2 as soon as the export
-    // positions' targets are written (sweep part 1), hiding the exchange
-    // behind the interior sweep and the epilogue. Joined via ev_halo at
-    // the next iteration's top.
-    auto issue_halo1a = [&](const i64 *commArr, i64 *dst) {
+    // Overlap mode: run the NEXT iteration's halo #1a on stream2/comm2 as
+    // soon as the export positions' targets are written (sweep part 1),
+    // hiding the exchange behind the interior sweep and the epilogue.
+    // Joined via ev_halo at the next iteration's top.
+    if (p > 1 && e->overlap) {
         HIP_CHECK(hipEventRecord(e->ev_p1, st));
         HIP_CHECK(hipStreamWaitEvent(e->stream2, e->ev_p1, 0));
-        k8_gather_comms<<<grid_for(std::max<i64>(e->ssz, 1)), 256, 0,
-                          e->stream2>>>(e->ssz, e->d_svdata_int, commArr,
-                                        e->d_scdata);
-        rccl_alltoallv(e, e->d_scdata, e->send_off.data(), dst,
-                       e->recv_off.data(), 8, ncclInt64, 8,
-                       e->comm2 ? e->comm2 : e->comm, e->stream2);
+        run_halo1a(e, e->d_curr, e->d_gc[0], e->comm2 ? e->comm2 : e->comm,
+                   e->stream2);
         HIP_CHECK(hipEventRecord(e->ev_halo, e->stream2));
-    };
-    if (p > 1 && e->overlap) issue_halo1a(e->d_curr, e->d_gc[0]);
+    }
 
     std::vector<hipEvent_t> sweep_ev;
     for (;;) {
@@ -2114,12 +2324,7 @@ double mv_engine_run(mv_engine *e, double lower, double thresh,
             if (e->overlap) {
                 HIP_CHECK(hipStreamWaitEvent(st, e->ev_halo, 0));
             } else {
-                k8_gather_comms<<<grid_for(std::max<i64>(e->ssz, 1)), 256, 0,
-                                  st>>>(e->ssz, e->d_svdata_int, d_curr,
-                                        e->d_scdata);
-                rccl_alltoallv(e, e->d_scdata, e->send_off.data(), gc,
-                               e->recv_off.data(), 8, ncclInt64, 8, e->comm,
-                               e->stream);
+                run_halo1a(e, d_curr, gc, e->comm, st);
             }
 
             // ---- needed remote communities (dspl.hpp:670-700) ----
@@ -2151,7 +2356,7 @@ double mv_engine_run(mv_engine *e, double lower, double thresh,
             }
             HIP_CHECK(hipMemsetAsync(e->d_count, 0, 8, st));
             k_filter_remote<<<grid_for(std::max<i64>(e->nghost, 1)), 256, 0,
-                              st>>>(e->nghost, gc, /*shift*/ 0,
+                              st>>>(e->nghost, e->d_ghost_labels, /*shift*/ 0,
                                     e->base, e->bound, e->d_cand, e->d_count);
             k_filter_remote<<<grid_for(lnv), 256, 0, st>>>(
                 lnv, d_curr, /*shift*/ 32, e->base, e->bound, e->d_cand,
@@ -2169,10 +2374,7 @@ double mv_engine_run(mv_engine *e, double lower, double thresh,
             (void)hipcub::DeviceSelect::Unique(e->d_cub_tmp, tb, e->d_cand_sorted,
                                          e->d_rc_ids, d_nrc, (int64_t)ncand, st);
             HIP_CHECK(hipMemcpyAsync(&nrc, d_nrc, 8, hipMemcpyDeviceToHost, st));
-            // ghost communities: labels -> handles for the sweep
-            k_labels_to_handles<<<grid_for(std::max<i64>(e->nghost, 1)), 256,
-                                  0, st>>>(e->nghost, e->base, e->bound,
-                                           e->d_sigma_inv, gc);
+            // (labels -> handles conversion happened inside run_halo1a)
             HIP_CHECK(hipStreamSynchronize(st));
 
             // ---- halo #1b/#1c/#1d: request (size,degree) of those
@@ -2336,11 +2538,15 @@ double mv_engine_run(mv_engine *e, double lower, double thresh,
             }
         };
         if (p > 1 && e->overlap) {
-            // exports first, then launch the NEXT iteration's #1a off their
-            // targets while the interior positions sweep
+            // exports first; the interior launch goes in BEFORE the halo's
+            // host-side work so it runs under the exchange
             sweep_range(0, e->nexp);
-            issue_halo1a(d_target, e->d_gc[gcur ^ 1]);
+            HIP_CHECK(hipEventRecord(e->ev_p1, st));
             sweep_range(e->nexp, lnv);
+            HIP_CHECK(hipStreamWaitEvent(e->stream2, e->ev_p1, 0));
+            run_halo1a(e, d_target, e->d_gc[gcur ^ 1],
+                       e->comm2 ? e->comm2 : e->comm, e->stream2);
+            HIP_CHECK(hipEventRecord(e->ev_halo, e->stream2));
         } else {
             sweep_range(e->nhi, lnv);
         }

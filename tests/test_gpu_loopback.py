@@ -115,6 +115,19 @@ def test_loopback_parity_no_overlap_p4():
         del os.environ["MV_NO_OVERLAP"]
 
 
+def test_loopback_parity_no_delta_p4():
+    """Full-resend #1a (MV_NO_DELTA, the reference's wire behavior) must
+    produce the same bits as the delta-compacted default."""
+    os.environ["MV_NO_DELTA"] = "1"
+    try:
+        pins = json.load(open(GOLDEN))
+        pin = pins["rgg_n16384_p4_unit"]
+        results = _run_loopback(16384, 4, unit=True)
+        _check_vs_pin(results, pin, 4, exact_mod=True)
+    finally:
+        del os.environ["MV_NO_DELTA"]
+
+
 def test_loopback_deterministic_p4():
     """Two identical p=4 loopback runs agree bit-for-bit (per-sender
     in-order delta application — run-to-run determinism at nranks > 2)."""
