@@ -186,15 +186,15 @@ __global__ void k_partial_sum2(i64 n, F f, double *__restrict__ out2) {
 }
 
 // ---- K3: community iota (dspl.hpp:132-149): internal slot k holds the
-// LABEL of the vertex it represents ----
+// LABEL of the vertex it represents (multi-rank arrays carry labels) ----
 __global__ void k3_init_comm(i64 lnv, i64 base,
                              const unsigned *__restrict__ sigma,
                              i64 *__restrict__ curr, i64 *__restrict__ past) {
     for (i64 k = blockIdx.x * (i64)blockDim.x + threadIdx.x; k < lnv;
          k += (i64)gridDim.x * blockDim.x) {
-        const i64 h = (((i64)sigma[k] + base) << 32) | k; // handle
-        curr[k] = h;
-        past[k] = h;
+        const i64 l = (i64)sigma[k] + base;
+        curr[k] = l;
+        past[k] = l;
     }
 }
 
@@ -306,18 +306,40 @@ __global__ void k_to_internal(i64 n, const i64 *__restrict__ gids, i64 base,
         out[k] = sigma_inv[gids[k] - base];
 }
 
-// ghost communities arrive as LABELS; convert to handles (out-of-place:
-// src stays the persistent label image the delta scatter maintains)
-__global__ void k_labels_to_handles(i64 n, i64 base, i64 bound,
-                                    const unsigned *__restrict__ sigma_inv,
-                                    const i64 *__restrict__ src,
-                                    i64 *__restrict__ dst) {
+// ---- the u32 community VIEW (multi-rank sweeps) ----
+// Persistent community arrays carry LABELS (the reference's community ids,
+// global vertex ids). Each iteration builds a u32 view: a local community
+// is its home vertex's internal SLOT (< lnv, indexes cinfo/cupd directly),
+// a remote one is lnv + its index in the sorted rc_ids array (indexes
+// rc_info/rcu directly). The per-edge community gather halves to 4 B and
+// the gain scan's per-candidate binary search disappears; only the
+// tie-break needs label ORDER, fetched lazily (see k4_sweep_p1's note).
+__global__ void k_build_view(i64 n, const i64 *__restrict__ labels, i64 base,
+                             i64 bound, i64 lnv,
+                             const unsigned *__restrict__ sigma_inv,
+                             const i64 *__restrict__ rc_ids, i64 nrc,
+                             unsigned *__restrict__ view) {
     for (i64 k = blockIdx.x * (i64)blockDim.x + threadIdx.x; k < n;
          k += (i64)gridDim.x * blockDim.x) {
-        const i64 c = src[k];
-        dst[k] = (c >= base && c < bound)
-                     ? (c << 32) | (i64)sigma_inv[c - base]
-                     : c << 32;
+        const i64 l = labels[k];
+        view[k] = (l >= base && l < bound)
+                      ? sigma_inv[l - base]
+                      : (unsigned)(lnv + clamp0(dev_bsearch(rc_ids, nrc, l)));
+    }
+}
+
+// targets come out of the sweep as views; persist them as labels
+__global__ void k_view_to_labels(i64 s0, i64 s1,
+                                 const unsigned *__restrict__ perm,
+                                 const unsigned *__restrict__ vtarget,
+                                 const unsigned *__restrict__ sigma, i64 base,
+                                 const i64 *__restrict__ rc_ids, i64 lnv,
+                                 i64 *__restrict__ out) {
+    for (i64 s = s0 + blockIdx.x * (i64)blockDim.x + threadIdx.x; s < s1;
+         s += (i64)gridDim.x * blockDim.x) {
+        const i64 i = perm[s];
+        const unsigned v = vtarget[i];
+        out[i] = (v < lnv) ? base + (i64)sigma[v] : rc_ids[v - lnv];
     }
 }
 
@@ -355,13 +377,13 @@ __global__ void k_scatter_deltas(i64 n, const unsigned *__restrict__ idx,
         ghost_labels[seg_base + idx[k]] = lab[k];
 }
 
-// ---- K8: scdata gather (dspl.hpp:559-571) ----
+// ---- K8: scdata gather (dspl.hpp:559-571); labels on the wire ----
 __global__ void k8_gather_comms(i64 n, const unsigned *__restrict__ svdata_int,
                                 const i64 *__restrict__ currComm,
                                 i64 *__restrict__ out) {
     for (i64 k = blockIdx.x * (i64)blockDim.x + threadIdx.x; k < n;
          k += (i64)gridDim.x * blockDim.x)
-        out[k] = currComm[svdata_int[k]] >> 32; // labels on the wire
+        out[k] = currComm[svdata_int[k]];
 }
 
 // ---- candidate remote communities (dspl.hpp:670-700) ----
@@ -458,13 +480,13 @@ __global__ void k67_apply_and_partials(i64 lnv, Cinfo *__restrict__ cupd,
     }
 }
 
-// de-permute an internal-ordered array into label order (trace only)
+// de-permute an internal-ordered label array into original order (trace)
 __global__ void k_depermute(i64 lnv, const unsigned *__restrict__ sigma_inv,
                             const i64 *__restrict__ in,
                             i64 *__restrict__ out) {
     for (i64 v = blockIdx.x * (i64)blockDim.x + threadIdx.x; v < lnv;
          v += (i64)gridDim.x * blockDim.x)
-        out[v] = in[sigma_inv[v]] >> 32; // handle -> label
+        out[v] = in[sigma_inv[v]];
 }
 
 // ---- K4: THE sweep (distExecuteLouvainIteration, dspl.hpp:276-405) ----
@@ -475,49 +497,59 @@ __global__ void k_depermute(i64 lnv, const unsigned *__restrict__ sigma_inv,
 // (only populated while communities are still fine-grained; L2-resident).
 // The current community's accumulator is a register (the reference's
 // counter[0], dspl.hpp:312-318).
+// ---- K4 multi-rank sweep over the u32 VIEW (see k_build_view) ----
+// Same structure as k4_sweep_p1: per-lane LDS slot arrays + spill, edges
+// in chunks of 8, lazy tie-break labels. A community value < lnv is a
+// local internal slot (cinfo/cupd index); >= lnv is lnv + rc index
+// (rc_info/rcu index). Labels for the tie-break / singleton guard come
+// from sigma (+base) or rc_ids, fetched only on gain ties (dspl.hpp:
+// 174-228 order semantics preserved bit-for-bit).
 template <int SLOTS, bool UNIT>
-__global__ __launch_bounds__(256) void k4_sweep(
-    i64 s_begin, i64 s_end, i64 lnv, i64 base, i64 bound,
+__global__ __launch_bounds__(256) void k4_sweep_mr(
+    i64 s_begin, i64 s_end, i64 lnv, i64 base,
     const unsigned *__restrict__ perm,
     const unsigned *__restrict__ deg_int, const i64 *__restrict__ chunk_off,
     const int *__restrict__ sell_tidx, const double *__restrict__ sell_w,
-    const i64 *__restrict__ currComm, const i64 *__restrict__ ghost_comm,
+    const unsigned *__restrict__ vcurr, const unsigned *__restrict__ vghost,
     const double *__restrict__ vDegree, const unsigned *__restrict__ sigma,
     const Cinfo *__restrict__ cinfo, Cinfo *__restrict__ cupd,
-    const i64 *__restrict__ rc_ids, i64 nrc,
+    const i64 *__restrict__ rc_ids,
     const Info16 *__restrict__ rc_info, Info16 *__restrict__ rcu,
-    double constant, i64 *__restrict__ targetComm,
-    double *__restrict__ clusterWeight, i64 *__restrict__ spill_keys,
-    double *__restrict__ spill_acc, const i64 *__restrict__ spill_off,
-    int dbg_skip) {
+    double constant, unsigned *__restrict__ vtarget,
+    double *__restrict__ clusterWeight, unsigned *__restrict__ spill_keys,
+    double *__restrict__ spill_acc, const i64 *__restrict__ spill_off) {
     extern __shared__ __attribute__((aligned(16))) char smem[];
-    i64 *skey = reinterpret_cast<i64 *>(smem);
-    double *sacc = reinterpret_cast<double *>(smem + sizeof(i64) * SLOTS * blockDim.x);
+    double *sacc = reinterpret_cast<double *>(smem);
+    unsigned *skey = reinterpret_cast<unsigned *>(
+        smem + sizeof(double) * SLOTS * blockDim.x);
     const int tid = threadIdx.x;
     const i64 gthread = blockIdx.x * (i64)blockDim.x + threadIdx.x;
     const i64 stride = (i64)gridDim.x * blockDim.x;
-    i64 *myspill_k = spill_keys + spill_off[gthread];
+    unsigned *myspill_k = spill_keys + spill_off[gthread];
     double *myspill_a = spill_acc + spill_off[gthread];
+
+    // lazy label fetch (tie-break order only, dspl.hpp:214-225)
+    auto label_of = [&](unsigned v) -> i64 {
+        return (v < lnv) ? base + (i64)sigma[v] : rc_ids[clamp0(v - lnv)];
+    };
 
     for (i64 s = s_begin + gthread; s < s_end; s += stride) {
         const i64 i = perm[s];          // internal vertex index
         const int deg = (int)deg_int[i];
         const i64 ebase = chunk_off[s >> 6] + (s & 63);
-        const i64 cc = currComm[i]; // handle
-        const i64 cl = h_label(cc);
-        const bool cc_local = (cl >= base && cl < bound);
+        const unsigned cc = vcurr[i];
         double ccDeg;
         i64 ccSize;
-        if (cc_local) { // dspl.hpp:296-307
-            const Cinfo c = cinfo[h_slot(cc)];
+        if (cc < lnv) { // dspl.hpp:296-307
+            const Cinfo c = cinfo[cc];
             ccDeg = c.degree;
             ccSize = c.size;
         } else {
-            const i64 q = clamp0(dev_bsearch(rc_ids, nrc, cl));
-            ccDeg = rc_info[q].degree;
-            ccSize = rc_info[q].size;
+            const Info16 c = rc_info[cc - lnv];
+            ccDeg = c.degree;
+            ccSize = c.size;
         }
-        i64 target;
+        unsigned target;
         if (deg == 0) {
             target = cc;          // dspl.hpp:323-324
             clusterWeight[i] = 0; // K5 semantics (dspl.hpp:481-482)
@@ -533,7 +565,7 @@ __global__ __launch_bounds__(256) void k4_sweep(
             for (int k0 = 0; k0 < deg; k0 += CH) {
                 const int m = min(CH, deg - k0);
                 i64 tb[CH];
-                i64 cb[CH];
+                unsigned cb[CH];
                 double wb[CH];
 #pragma unroll
                 for (int j = 0; j < CH; j++) {
@@ -544,13 +576,13 @@ __global__ __launch_bounds__(256) void k4_sweep(
                 }
 #pragma unroll
                 for (int j = 0; j < CH; j++)
-                    cb[j] = (tb[j] < lnv) ? currComm[tb[j]]
-                                          : ghost_comm[clamp0(tb[j] - lnv)];
+                    cb[j] = (tb[j] < lnv) ? vcurr[tb[j]]
+                                          : vghost[clamp0(tb[j] - lnv)];
                 for (int j = 0; j < m; j++) {
                     const i64 tidx = tb[j];
                     const double w = UNIT ? 1.0 : wb[j];
                     if (tidx == i) selfLoop += w; // dspl.hpp:247-248
-                    const i64 tcomm = cb[j];
+                    const unsigned tcomm = cb[j];
                     if (tcomm == cc) { c0 += w; continue; }
                     bool found = false;
                     for (int t = 0; t < ns; t++) {
@@ -567,7 +599,6 @@ __global__ __launch_bounds__(256) void k4_sweep(
                         ns++;
                         continue;
                     }
-                    if (dbg_skip & 4) continue; // DEBUG: drop spill
                     for (int t = 0; t < nspill; t++) {
                         if (myspill_k[t] == tcomm) {
                             myspill_a[t] += w;
@@ -584,131 +615,136 @@ __global__ __launch_bounds__(256) void k4_sweep(
             }
             clusterWeight[i] = c0; // dspl.hpp:318 onto the K5-zeroed value
 
-            // distGetMaxIndex (dspl.hpp:174-228)
+            // distGetMaxIndex (dspl.hpp:174-228); labels fetched lazily
             const double vdeg = vDegree[i];
             const double eix = c0 - selfLoop;
             const double ax = ccDeg - vdeg;
             double maxGain = 0.0;
-            i64 maxIndex = cc, maxSize = ccSize;
-            const int tot = (dbg_skip & 2) ? 0 : ns + nspill;
+            unsigned maxIndex = cc;
+            i64 maxSize = ccSize;
+            const int tot = ns + nspill;
             for (int t = 0; t < tot; t++) {
-                const i64 y = (t < ns) ? skey[t * blockDim.x + tid]
-                                       : myspill_k[t - ns];
+                const unsigned y = (t < ns) ? skey[t * blockDim.x + tid]
+                                            : myspill_k[t - ns];
                 const double eiy = (t < ns) ? sacc[t * blockDim.x + tid]
                                             : myspill_a[t - ns];
-                if (dbg_skip & 16) { continue; } // DEBUG: read-only scan
-                const i64 yl = h_label(y);
                 double ay;
                 i64 ysz;
-                if (dbg_skip & 8) { ay = 0.0; ysz = 1; } // DEBUG: no lookup
-                else if (yl >= base && yl < bound) {
-                    const Cinfo c = cinfo[h_slot(y)];
+                if (y < lnv) {
+                    const Cinfo c = cinfo[y];
                     ay = c.degree;
                     ysz = c.size;
                 } else {
-                    const i64 q = clamp0(dev_bsearch(rc_ids, nrc, yl));
-                    ay = rc_info[q].degree;
-                    ysz = rc_info[q].size;
+                    const Info16 c = rc_info[y - lnv];
+                    ay = c.degree;
+                    ysz = c.size;
                 }
                 const double curGain =
                     2.0 * (eiy - eix) - 2.0 * vdeg * (ay - ax) * constant; // :212
-                if (curGain > maxGain ||
-                    (curGain == maxGain && curGain != 0.0 && y < maxIndex)) {
+                if (curGain > maxGain) {
                     maxGain = curGain;
                     maxIndex = y;
                     maxSize = ysz;
+                } else if (curGain == maxGain && curGain != 0.0) {
+                    // real branch, not a select: the label loads must not
+                    // be folded into the gain-compare dataflow (ROCm 7.2
+                    // if-conversion miscompile)
+                    if (label_of(y) < label_of(maxIndex)) {
+                        maxIndex = y;
+                        maxSize = ysz;
+                    }
                 }
             }
-            if (maxSize == 1 && ccSize == 1 && maxIndex > cc) // :224-225
-                maxIndex = cc;
+            if (maxSize == 1 && ccSize == 1 && maxIndex != cc) { // :224-225
+                if (label_of(maxIndex) > label_of(cc)) maxIndex = cc;
+            }
             target = maxIndex;
         }
 
-        if (target != cc && !(dbg_skip & 1)) { // 4-case updates (dspl.hpp:331-399)
+        if (target != cc) { // 4-case updates (dspl.hpp:331-399)
             const double vdeg = vDegree[i];
-            if (cc_local) {
-                Cinfo *u = &cupd[h_slot(cc)];
+            if (cc < lnv) {
+                Cinfo *u = &cupd[cc];
                 atomicAdd(&u->degree, -vdeg);
                 atomic_add_i64(&u->size, -1);
             } else {
-                const i64 q = dev_bsearch(rc_ids, nrc, cl);
-                atomicAdd(&rcu[q].degree, -vdeg);
-                atomic_add_i64(&rcu[q].size, -1);
+                Info16 *u = &rcu[cc - lnv];
+                atomicAdd(&u->degree, -vdeg);
+                atomic_add_i64(&u->size, -1);
             }
-            const i64 tl = h_label(target);
-            if (tl >= base && tl < bound) {
-                Cinfo *u = &cupd[h_slot(target)];
+            if (target < lnv) {
+                Cinfo *u = &cupd[target];
                 atomicAdd(&u->degree, vdeg);
                 atomic_add_i64(&u->size, 1);
             } else {
-                const i64 q = dev_bsearch(rc_ids, nrc, tl);
-                atomicAdd(&rcu[q].degree, vdeg);
-                atomic_add_i64(&rcu[q].size, 1);
+                Info16 *u = &rcu[target - lnv];
+                atomicAdd(&u->degree, vdeg);
+                atomic_add_i64(&u->size, 1);
             }
         }
-        targetComm[i] = target; // dspl.hpp:404 (handle)
+        vtarget[i] = target; // dspl.hpp:404 (view; persisted as a label)
     }
 }
 
-// ---- K4 high-degree path (wave-per-vertex) ----
+// ---- K4 high-degree path (wave-per-vertex), multi-rank VIEW mode ----
 // For skewed graphs (Orkut-class hubs), one WAVE processes one vertex:
 // lanes stride the CSR row (coalesced 8-B tails), aggregate into a
-// per-vertex open-addressed hash in HBM (L2-resident; keys CASed from -1,
-// weights atomicAdd), then a wave-reduce argmax applies the reference
-// tie-break as a total order on (gain, label) — order-independent, so the
-// reduce tree reproduces dspl.hpp:214-215 exactly. Unit-weight graphs
-// only: their sums are integer-exact under any accumulation order; -w
-// skewed graphs take the serial lane path instead (edge-order bit parity).
-template <bool UNIT>
-__global__ __launch_bounds__(256) void k4_sweep_hi(
+// per-vertex open-addressed hash in HBM (L2-resident; u32 view keys CASed
+// from ~0, weights atomicAdd), then a wave-reduce argmax applies the
+// reference tie-break as a total order on (gain, label<<32|view) — order-
+// independent, so the reduce tree reproduces dspl.hpp:214-215 exactly.
+// Unit-weight graphs only: their sums are integer-exact under any
+// accumulation order; -w skewed graphs take the serial lane path instead
+// (edge-order bit parity).
+__global__ __launch_bounds__(256) void k4_sweep_hi_mr(
     i64 nhi, i64 lnv, i64 base, i64 bound, const unsigned *__restrict__ perm,
     const unsigned *__restrict__ deg_int, const unsigned *__restrict__ sigma,
     const unsigned *__restrict__ sigma_inv, const i64 *__restrict__ xadj,
-    const i64 *__restrict__ tails, const double *__restrict__ ew,
-    const i64 *__restrict__ ghosts, i64 nghost,
-    const i64 *__restrict__ currComm, const i64 *__restrict__ ghost_comm,
+    const i64 *__restrict__ tails, const i64 *__restrict__ ghosts, i64 nghost,
+    const unsigned *__restrict__ vcurr, const unsigned *__restrict__ vghost,
     const double *__restrict__ vDegree, const Cinfo *__restrict__ cinfo,
-    Cinfo *__restrict__ cupd, const i64 *__restrict__ rc_ids, i64 nrc,
+    Cinfo *__restrict__ cupd, const i64 *__restrict__ rc_ids,
     const Info16 *__restrict__ rc_info, Info16 *__restrict__ rcu,
-    double constant, i64 *__restrict__ targetComm,
+    double constant, unsigned *__restrict__ vtarget,
     double *__restrict__ clusterWeight, const i64 *__restrict__ hash_off,
     i64 *__restrict__ hkeys, double *__restrict__ hacc) {
     const int lane = threadIdx.x & 63;
     const int wid = threadIdx.x >> 6;
     const int wpb = blockDim.x >> 6;
+    auto label_of = [&](unsigned v) -> i64 {
+        return (v < lnv) ? base + (i64)sigma[v] : rc_ids[clamp0(v - lnv)];
+    };
     for (i64 s = (i64)blockIdx.x * wpb + wid; s < nhi;
          s += (i64)gridDim.x * wpb) {
         const i64 i = perm[s];
         const i64 v = sigma[i];
         const int deg = (int)deg_int[i];
         const i64 e0 = xadj[v];
-        const i64 cc = currComm[i]; // handle
-        const i64 cl = h_label(cc);
-        const bool cc_local = (cl >= base && cl < bound);
+        const unsigned cc = vcurr[i];
         const i64 hoff = hash_off[s];
         const i64 cap = hash_off[s + 1] - hoff; // power of two
         const i64 vglobal = v + base;
         double ccDeg;
         i64 ccSize;
-        if (cc_local) {
-            const Cinfo c = cinfo[h_slot(cc)];
+        if (cc < lnv) {
+            const Cinfo c = cinfo[cc];
             ccDeg = c.degree;
             ccSize = c.size;
         } else {
-            const i64 q = clamp0(dev_bsearch(rc_ids, nrc, cl));
-            ccDeg = rc_info[q].degree;
-            ccSize = rc_info[q].size;
+            const Info16 c = rc_info[cc - lnv];
+            ccDeg = c.degree;
+            ccSize = c.size;
         }
         double c0 = 0.0, selfLoop = 0.0;
         for (int k = lane; k < deg; k += 64) {
             const i64 tail = tails[e0 + k];
-            const double w = UNIT ? 1.0 : ew[e0 + k];
+            const double w = 1.0; // hub path is unit-only
             if (tail == vglobal) selfLoop += w;
             const i64 ti = (tail >= base && tail < bound)
                                ? (i64)sigma_inv[tail - base]
                                : lnv + dev_lower_bound(ghosts, nghost, tail);
-            const i64 tcomm = (ti < lnv) ? currComm[ti]
-                                         : ghost_comm[ti - lnv];
+            const unsigned tcomm = (ti < lnv) ? vcurr[ti]
+                                              : vghost[ti - lnv];
             if (tcomm == cc) { c0 += w; continue; }
             i64 pos = (i64)(((uint64_t)tcomm * 0x9E3779B97F4A7C15ull) >> 32) &
                       (cap - 1);
@@ -716,7 +752,7 @@ __global__ __launch_bounds__(256) void k4_sweep_hi(
                 const i64 prev = (i64)atomicCAS(
                     (unsigned long long *)&hkeys[hoff + pos],
                     (unsigned long long)(-1ll), (unsigned long long)tcomm);
-                if (prev == -1 || prev == tcomm) {
+                if (prev == -1 || prev == (i64)tcomm) {
                     atomicAdd(&hacc[hoff + pos], w);
                     break;
                 }
@@ -734,31 +770,32 @@ __global__ __launch_bounds__(256) void k4_sweep_hi(
         const double eix = c0 - selfLoop;
         const double ax = ccDeg - vdeg;
         // per-lane best over hash slots, then wave-reduce under the total
-        // order (gain desc, label asc); gain <= 0 maps to the neutral
-        // element so it can never win (dspl.hpp:214-215 semantics)
+        // order (gain desc, (label, view) asc); gain <= 0 maps to the
+        // neutral element so it can never win (dspl.hpp:214-215)
         double bg = 0.0;
-        i64 bl = INT64_MAX, bs = 0;
+        i64 bl = INT64_MAX, bs = 0; // bl = (label << 32) | view
         for (i64 t = lane; t < cap; t += 64) {
-            const i64 y = hkeys[hoff + t];
-            if (y == -1) continue;
+            const i64 yk = hkeys[hoff + t];
+            if (yk == -1) continue;
+            const unsigned y = (unsigned)yk;
             const double eiy = hacc[hoff + t];
-            const i64 yl = h_label(y);
             double ay;
             i64 ysz;
-            if (yl >= base && yl < bound) {
-                const Cinfo c = cinfo[h_slot(y)];
+            if (y < lnv) {
+                const Cinfo c = cinfo[y];
                 ay = c.degree;
                 ysz = c.size;
             } else {
-                const i64 q = clamp0(dev_bsearch(rc_ids, nrc, yl));
-                ay = rc_info[q].degree;
-                ysz = rc_info[q].size;
+                const Info16 c = rc_info[y - lnv];
+                ay = c.degree;
+                ysz = c.size;
             }
             const double g =
                 2.0 * (eiy - eix) - 2.0 * vdeg * (ay - ax) * constant;
-            if (g > bg || (g == bg && g != 0.0 && y < bl)) {
+            const i64 yh = (label_of(y) << 32) | (i64)y;
+            if (g > bg || (g == bg && g != 0.0 && yh < bl)) {
                 bg = g;
-                bl = y;
+                bl = yh;
                 bs = ysz;
             }
         }
@@ -773,8 +810,12 @@ __global__ __launch_bounds__(256) void k4_sweep_hi(
             }
         }
         if (lane == 0) {
-            i64 target = (bl == INT64_MAX) ? cc : bl;
-            if (bs == 1 && ccSize == 1 && target > cc) target = cc; // :224-225
+            unsigned target =
+                (bl == INT64_MAX) ? cc : (unsigned)(bl & 0xffffffffu);
+            const i64 tLabel =
+                (bl == INT64_MAX) ? label_of(cc) : (bl >> 32);
+            if (bs == 1 && ccSize == 1 && tLabel > label_of(cc))
+                target = cc; // :224-225
             if (deg == 0) {
                 clusterWeight[i] = 0;
                 target = cc;
@@ -782,27 +823,26 @@ __global__ __launch_bounds__(256) void k4_sweep_hi(
                 clusterWeight[i] = c0;
             }
             if (target != cc) {
-                if (cc_local) {
-                    Cinfo *u = &cupd[h_slot(cc)];
+                if (cc < lnv) {
+                    Cinfo *u = &cupd[cc];
                     atomicAdd(&u->degree, -vdeg);
                     atomic_add_i64(&u->size, -1);
                 } else {
-                    const i64 q = dev_bsearch(rc_ids, nrc, cl);
-                    atomicAdd(&rcu[q].degree, -vdeg);
-                    atomic_add_i64(&rcu[q].size, -1);
+                    Info16 *u = &rcu[cc - lnv];
+                    atomicAdd(&u->degree, -vdeg);
+                    atomic_add_i64(&u->size, -1);
                 }
-                const i64 tl2 = h_label(target);
-                if (tl2 >= base && tl2 < bound) {
-                    Cinfo *u = &cupd[h_slot(target)];
+                if (target < lnv) {
+                    Cinfo *u = &cupd[target];
                     atomicAdd(&u->degree, vdeg);
                     atomic_add_i64(&u->size, 1);
                 } else {
-                    const i64 q = dev_bsearch(rc_ids, nrc, tl2);
-                    atomicAdd(&rcu[q].degree, vdeg);
-                    atomic_add_i64(&rcu[q].size, 1);
+                    Info16 *u = &rcu[target - lnv];
+                    atomicAdd(&u->degree, vdeg);
+                    atomic_add_i64(&u->size, 1);
                 }
             }
-            targetComm[i] = target;
+            vtarget[i] = target;
         }
     }
 }
@@ -821,17 +861,23 @@ __global__ __launch_bounds__(256) void k4_sweep_hi(
 // random cinfo line. The singleton guard (dspl.hpp:224-225) is the final
 // maxIndex > cc check. Gains carry the identical bits: 2.0*(eiy-0.0) and
 // (ay-0.0) round exactly like the reference's expressions.
+// Multi-rank iteration-1 streaming specialization, VIEW mode (see
+// k4_sweep_iter1_p1's header note). currComm is the identity, so a LOCAL
+// candidate's view is its tail index and its ay is vDegree[tidx]; a ghost
+// candidate's view comes from vghost (already lnv + rc index — the old
+// per-edge rc binary search disappears). Labels for the final singleton
+// guard come batched from sigma / the sorted ghost id list.
 template <bool UNIT>
-__global__ __launch_bounds__(256) void k4_sweep_iter1(
-    i64 s_begin, i64 s_end, i64 lnv, i64 base, i64 bound,
+__global__ __launch_bounds__(256) void k4_sweep_iter1_mr(
+    i64 s_begin, i64 s_end, i64 lnv, i64 base,
     const unsigned *__restrict__ perm,
     const unsigned *__restrict__ deg_int, const i64 *__restrict__ chunk_off,
     const int *__restrict__ sell_tidx, const double *__restrict__ sell_w,
-    const i64 *__restrict__ currComm, const i64 *__restrict__ ghost_comm,
-    const double *__restrict__ vDegree, Cinfo *__restrict__ cupd,
-    const i64 *__restrict__ rc_ids, i64 nrc,
+    const unsigned *__restrict__ vghost, const i64 *__restrict__ ghosts,
+    const double *__restrict__ vDegree, const unsigned *__restrict__ sigma,
+    Cinfo *__restrict__ cupd,
     const Info16 *__restrict__ rc_info, Info16 *__restrict__ rcu,
-    double constant, i64 *__restrict__ targetComm,
+    double constant, unsigned *__restrict__ vtarget,
     double *__restrict__ clusterWeight) {
     const i64 gthread = blockIdx.x * (i64)blockDim.x + threadIdx.x;
     const i64 stride = (i64)gridDim.x * blockDim.x;
@@ -839,19 +885,22 @@ __global__ __launch_bounds__(256) void k4_sweep_iter1(
         const i64 i = perm[s];
         const int deg = (int)deg_int[i];
         const i64 ebase = chunk_off[s >> 6] + (s & 63);
-        const i64 cc = currComm[i]; // own label
+        const unsigned cc = (unsigned)i; // own slot (identity)
+        const i64 ccLabel = base + (i64)sigma[i];
         const double vdeg = vDegree[i];
         double c0 = 0.0;
         double maxGain = 0.0;
-        i64 maxIndex = cc;
+        unsigned maxIndex = cc;
+        i64 maxLabel = ccLabel;
         i64 prev = INT64_MIN;
+        unsigned pend_view = 0;
         i64 pend_label = 0;
         double eiy = 0.0, pend_ay = 0.0;
         bool pend = false;
         constexpr int CH = 8;
         for (int k0 = 0; k0 < deg; k0 += CH) {
             const int m = min(CH, deg - k0);
-            i64 tb[CH], cb[CH];
+            i64 tb[CH], lb[CH];
             double wb[CH], vb[CH];
 #pragma unroll
             for (int j = 0; j < CH; j++) {
@@ -861,8 +910,8 @@ __global__ __launch_bounds__(256) void k4_sweep_iter1(
             }
 #pragma unroll
             for (int j = 0; j < CH; j++) {
-                cb[j] = (tb[j] < lnv) ? currComm[tb[j]]
-                                      : ghost_comm[clamp0(tb[j] - lnv)];
+                lb[j] = (tb[j] < lnv) ? base + (i64)sigma[tb[j]]
+                                      : ghosts[clamp0(tb[j] - lnv)];
                 vb[j] = (tb[j] < lnv) ? vDegree[tb[j]] : 0.0;
             }
             for (int j = 0; j < m; j++) {
@@ -873,41 +922,50 @@ __global__ __launch_bounds__(256) void k4_sweep_iter1(
                 if (pend) {
                     const double g =
                         2.0 * eiy - 2.0 * vdeg * pend_ay * constant;
-                    if (g > maxGain) { maxGain = g; maxIndex = pend_label; }
+                    if (g > maxGain) {
+                        maxGain = g;
+                        maxIndex = pend_view;
+                        maxLabel = pend_label;
+                    }
                 }
                 prev = tidx;
-                pend_label = cb[j]; // handle (label-ordered like the tails)
+                if (tidx < lnv) {
+                    pend_view = (unsigned)tidx; // candidate comm == tail
+                    pend_ay = vb[j];
+                } else {
+                    pend_view = vghost[tidx - lnv]; // lnv + rc index
+                    pend_ay = rc_info[clamp0((i64)pend_view - lnv)].degree;
+                }
+                pend_label = lb[j]; // label-ordered like the tails
                 eiy = w;
-                pend_ay = (tidx < lnv)
-                              ? vb[j]
-                              : rc_info[clamp0(dev_bsearch(
-                                    rc_ids, nrc, h_label(cb[j])))].degree;
                 pend = true;
             }
         }
         if (pend) {
             const double g = 2.0 * eiy - 2.0 * vdeg * pend_ay * constant;
-            if (g > maxGain) { maxGain = g; maxIndex = pend_label; }
+            if (g > maxGain) {
+                maxGain = g;
+                maxIndex = pend_view;
+                maxLabel = pend_label;
+            }
         }
-        if (maxIndex > cc) maxIndex = cc; // singleton guard (handle order
-                                          // == label order)
-        clusterWeight[i] = c0;            // dspl.hpp:318 (eix == 0)
-        if (maxIndex != cc) {             // cc is local at iteration 1
-            Cinfo *u = &cupd[h_slot(cc)];
+        if (maxLabel > ccLabel) maxIndex = cc; // singleton guard
+        clusterWeight[i] = c0;                 // dspl.hpp:318 (eix == 0)
+        if (maxIndex != cc) {                  // cc is local at iteration 1
+            Cinfo *u = &cupd[i];
             atomicAdd(&u->degree, -vdeg);
             atomic_add_i64(&u->size, -1);
-            const i64 tl = h_label(maxIndex);
-            if (tl >= base && tl < bound) {
-                Cinfo *t = &cupd[h_slot(maxIndex)];
+            if (maxIndex < lnv) {
+                Cinfo *t = &cupd[maxIndex];
                 atomicAdd(&t->degree, vdeg);
                 atomic_add_i64(&t->size, 1);
             } else {
-                const i64 q = dev_bsearch(rc_ids, nrc, tl);
-                atomicAdd(&rcu[q].degree, vdeg);
-                atomic_add_i64(&rcu[q].size, 1);
+                Info16 *t = &rcu[maxIndex - lnv];
+                atomicAdd(&t->degree, vdeg);
+                atomic_add_i64(&t->size, 1);
             }
         }
-        targetComm[i] = maxIndex; // handle
+        vtarget[i] = maxIndex;
     }
 }
 
@@ -1378,8 +1436,11 @@ struct mv_engine {
     // ghosts / halo
     i64 *d_ghosts = nullptr;     // sorted unique remote tails
     i64 nghost = 0;
-    i64 *d_gc[2] = {nullptr, nullptr}; // ghost communities (double-
-                                       // buffered for #1a overlap)
+    // u32 community views (multi-rank sweeps; rebuilt per iteration from
+    // the persistent LABEL arrays once rc_ids is known)
+    unsigned *d_vcurr = nullptr;   // lnv
+    unsigned *d_vghost = nullptr;  // nghost
+    unsigned *d_vtarget = nullptr; // lnv (persisted back as labels)
     i64 *d_svdata = nullptr;     // vertices peers want from me (global ids)
     unsigned *d_svdata_int = nullptr;
     i64 ssz = 0;
@@ -1559,8 +1620,8 @@ static void free_graph_state(mv_engine *e) {
                      (void **)&e->d_cupd, (void **)&e->d_partials,
                      (void **)&e->d_red, (void **)&e->d_count,
                      (void **)&e->d_bounds, (void **)&e->d_ghosts,
-                     (void **)&e->d_gc[0], (void **)&e->d_gc[1],
-                     (void **)&e->d_svdata,
+                     (void **)&e->d_vcurr, (void **)&e->d_vghost,
+                     (void **)&e->d_vtarget, (void **)&e->d_svdata,
                      (void **)&e->d_svdata_int, (void **)&e->d_scdata,
                      (void **)&e->d_cand, (void **)&e->d_cand_sorted,
                      (void **)&e->d_rc_ids, (void **)&e->d_rc_info,
@@ -1672,7 +1733,7 @@ int mv_engine_load_graph(mv_engine *e, const mv_graph *g) {
     HIP_CHECK(hipMalloc(&e->d_cupd, sizeof(Cinfo) * lnv));
     // never-null dummies for pointers that stay unused at nranks==1 but
     // may still be address-computed by if-converted loads
-    if (!e->d_gc[0]) HIP_CHECK(hipMalloc(&e->d_gc[0], 16));
+    if (!e->d_vghost) HIP_CHECK(hipMalloc(&e->d_vghost, 16));
     if (!e->d_rc_ids) HIP_CHECK(hipMalloc(&e->d_rc_ids, 16));
     if (!e->d_rc_info) HIP_CHECK(hipMalloc(&e->d_rc_info, sizeof(Info16)));
     if (!e->d_rcu) HIP_CHECK(hipMalloc(&e->d_rcu, sizeof(Info16)));
@@ -1996,11 +2057,12 @@ static void build_sell(mv_engine *e) {
 
 // Halo #1a (dspl.hpp:583-647) on (comm, stream): gather the exports'
 // communities, exchange them (full resend like the reference, or
-// delta-compacted), scatter into the persistent label image, and emit the
-// HANDLE image for the sweep into gc_dst. In overlap mode this runs on
-// stream2/comm2 concurrently with the interior sweep + epilogue.
-static void run_halo1a(mv_engine *e, const i64 *commArr, i64 *gc_dst,
-                       ncclComm_t comm, hipStream_t st2) {
+// delta-compacted), and scatter into the persistent label image
+// d_ghost_labels (the per-iteration u32 view is built from it once the
+// rc set is known). In overlap mode this runs on stream2/comm2
+// concurrently with the interior sweep + epilogue.
+static void run_halo1a(mv_engine *e, const i64 *commArr, ncclComm_t comm,
+                       hipStream_t st2) {
     const int p = e->nranks;
     k8_gather_comms<<<grid_for(std::max<i64>(e->ssz, 1)), 256, 0, st2>>>(
         e->ssz, e->d_svdata_int, commArr, e->d_scdata);
@@ -2113,9 +2175,6 @@ static void run_halo1a(mv_engine *e, const i64 *commArr, i64 *gc_dst,
                     e->d_ghost_labels);
         }
     }
-    k_labels_to_handles<<<grid_for(std::max<i64>(e->nghost, 1)), 256, 0,
-                          st2>>>(e->nghost, e->base, e->bound,
-                                 e->d_sigma_inv, e->d_ghost_labels, gc_dst);
 }
 
 #define PHASE(tag)                                                            \
@@ -2211,11 +2270,11 @@ double mv_engine_run(mv_engine *e, double lower, double thresh,
                 e->ssz, e->d_svdata, e->base, e->d_sigma_inv, e->d_svdata_int);
             HIP_CHECK(hipStreamSynchronize(st));
 
-            for (int b = 0; b < 2; b++) {
-                if (e->d_gc[b]) HIP_CHECK(hipFree(e->d_gc[b]));
-                HIP_CHECK(hipMalloc(&e->d_gc[b],
-                                    8 * std::max<i64>(e->nghost, 1)));
-            }
+            if (e->d_vghost) HIP_CHECK(hipFree(e->d_vghost));
+            HIP_CHECK(hipMalloc(&e->d_vghost,
+                                4 * std::max<i64>(e->nghost, 1)));
+            if (!e->d_vcurr) HIP_CHECK(hipMalloc(&e->d_vcurr, 4 * lnv));
+            if (!e->d_vtarget) HIP_CHECK(hipMalloc(&e->d_vtarget, 4 * lnv));
             if (e->d_scdata) HIP_CHECK(hipFree(e->d_scdata));
             HIP_CHECK(hipMalloc(&e->d_scdata, 8 * std::max<i64>(e->ssz, 1)));
 
@@ -2297,7 +2356,6 @@ double mv_engine_run(mv_engine *e, double lower, double thresh,
     i64 *d_curr = e->d_curr, *d_past = e->d_past, *d_target = e->d_target;
     double prevMod = lower, currMod = -1.0;
     int numIters = 0;
-    int gcur = 0; // which d_gc buffer THIS iteration's ghost comms live in
     std::vector<i64> rc_bounds(p + 1, 0), req_off(p + 1, 0);
 
     // Overlap mode: run the NEXT iteration's halo #1a on stream2/comm2 as
@@ -2307,8 +2365,7 @@ double mv_engine_run(mv_engine *e, double lower, double thresh,
     if (p > 1 && e->overlap) {
         HIP_CHECK(hipEventRecord(e->ev_p1, st));
         HIP_CHECK(hipStreamWaitEvent(e->stream2, e->ev_p1, 0));
-        run_halo1a(e, e->d_curr, e->d_gc[0], e->comm2 ? e->comm2 : e->comm,
-                   e->stream2);
+        run_halo1a(e, e->d_curr, e->comm2 ? e->comm2 : e->comm, e->stream2);
         HIP_CHECK(hipEventRecord(e->ev_halo, e->stream2));
     }
 
@@ -2317,14 +2374,13 @@ double mv_engine_run(mv_engine *e, double lower, double thresh,
         numIters++;
 
         i64 nrc = 0;
-        i64 *gc = e->d_gc[gcur];
         if (p > 1) {
             const auto t_h0 = std::chrono::steady_clock::now();
             // ---- halo #1a: ghost communities (dspl.hpp:583-647) ----
             if (e->overlap) {
                 HIP_CHECK(hipStreamWaitEvent(st, e->ev_halo, 0));
             } else {
-                run_halo1a(e, d_curr, gc, e->comm, st);
+                run_halo1a(e, d_curr, e->comm, st);
             }
 
             // ---- needed remote communities (dspl.hpp:670-700) ----
@@ -2359,7 +2415,7 @@ double mv_engine_run(mv_engine *e, double lower, double thresh,
                               st>>>(e->nghost, e->d_ghost_labels, /*shift*/ 0,
                                     e->base, e->bound, e->d_cand, e->d_count);
             k_filter_remote<<<grid_for(lnv), 256, 0, st>>>(
-                lnv, d_curr, /*shift*/ 32, e->base, e->bound, e->d_cand,
+                lnv, d_curr, /*shift*/ 0, e->base, e->bound, e->d_cand,
                 e->d_count);
             unsigned long long ncand = 0;
             HIP_CHECK(hipMemcpyAsync(&ncand, e->d_count, 8,
@@ -2413,6 +2469,14 @@ double mv_engine_run(mv_engine *e, double lower, double thresh,
             HIP_CHECK(hipMemsetAsync(e->d_rcu, 0,
                                      sizeof(Info16) * std::max<i64>(nrc, 1),
                                      st));
+            // u32 views for the sweep (slot / lnv + rc index encoding)
+            k_build_view<<<grid_for(lnv), 256, 0, st>>>(
+                lnv, d_curr, e->base, e->bound, lnv, e->d_sigma_inv,
+                e->d_rc_ids, nrc, e->d_vcurr);
+            k_build_view<<<grid_for(std::max<i64>(e->nghost, 1)), 256, 0,
+                           st>>>(e->nghost, e->d_ghost_labels, e->base,
+                                 e->bound, lnv, e->d_sigma_inv, e->d_rc_ids,
+                                 nrc, e->d_vghost);
             HIP_CHECK(hipStreamSynchronize(st));
             e->stats.halo_ms +=
                 std::chrono::duration<double, std::milli>(
@@ -2467,15 +2531,14 @@ double mv_engine_run(mv_engine *e, double lower, double thresh,
                         e->d_spill_off);
                 return;
             }
-            k4_sweep<S, decltype(unit_tag)::value>
-                <<<grid_for(s1 - s0, 256, 2048), 256, S * 256 * 16, st>>>(
-                    s0, s1, lnv, e->base, e->bound, e->d_perm, e->d_deg,
-                    e->d_chunk_off, e->d_sell_tidx, e->d_sell_w, d_curr,
-                    gc, e->d_vdeg, e->d_sigma, e->d_cinfo,
-                    e->d_cupd, e->d_rc_ids, nrc, e->d_rc_info, e->d_rcu,
-                    constant, d_target, e->d_cw, e->d_spill_k, e->d_spill_a,
-                    e->d_spill_off,
-                    getenv("MV_DBG_SKIP") ? atoi(getenv("MV_DBG_SKIP")) : 0);
+            k4_sweep_mr<S, decltype(unit_tag)::value>
+                <<<grid_for(s1 - s0, 256, 2048), 256, S * 256 * 12, st>>>(
+                    s0, s1, lnv, e->base, e->d_perm, e->d_deg,
+                    e->d_chunk_off, e->d_sell_tidx, e->d_sell_w, e->d_vcurr,
+                    e->d_vghost, e->d_vdeg, e->d_sigma, e->d_cinfo,
+                    e->d_cupd, e->d_rc_ids, e->d_rc_info, e->d_rcu,
+                    constant, e->d_vtarget, e->d_cw,
+                    (unsigned *)e->d_spill_k, e->d_spill_a, e->d_spill_off);
         };
         auto dispatch_slots = [&](auto unit_tag, i64 s0, i64 s1) {
             switch (slots) {
@@ -2495,12 +2558,12 @@ double mv_engine_run(mv_engine *e, double lower, double thresh,
                         e->d_cupd, constant, (unsigned *)d_target, e->d_cw);
                 return;
             }
-            k4_sweep_iter1<decltype(unit_tag)::value>
+            k4_sweep_iter1_mr<decltype(unit_tag)::value>
                 <<<grid_for(s1 - s0, 256, 2048), 256, 0, st>>>(
-                    s0, s1, lnv, e->base, e->bound, e->d_perm, e->d_deg,
-                    e->d_chunk_off, e->d_sell_tidx, e->d_sell_w, d_curr,
-                    gc, e->d_vdeg, e->d_cupd, e->d_rc_ids, nrc,
-                    e->d_rc_info, e->d_rcu, constant, d_target, e->d_cw);
+                    s0, s1, lnv, e->base, e->d_perm, e->d_deg,
+                    e->d_chunk_off, e->d_sell_tidx, e->d_sell_w, e->d_vghost,
+                    e->d_ghosts, e->d_vdeg, e->d_sigma, e->d_cupd,
+                    e->d_rc_info, e->d_rcu, constant, e->d_vtarget, e->d_cw);
         };
         if (e->nhi > 0) { // wave-per-vertex hubs (unit weights only)
             HIP_CHECK(hipMemsetAsync(e->d_hkeys, 0xFF, 8 * e->hash_total, st));
@@ -2514,13 +2577,13 @@ double mv_engine_run(mv_engine *e, double lower, double thresh,
                     e->d_cupd, constant, (unsigned *)d_target, e->d_cw,
                     e->d_hash_off, e->d_hkeys, e->d_hacc);
             else
-            k4_sweep_hi<true><<<grid_for(e->nhi * 64, 256, 2048), 256, 0,
-                                st>>>(
+            k4_sweep_hi_mr<<<grid_for(e->nhi * 64, 256, 2048), 256, 0,
+                             st>>>(
                 e->nhi, lnv, e->base, e->bound, e->d_perm, e->d_deg,
-                e->d_sigma, e->d_sigma_inv, e->d_xadj, e->d_tails, e->d_ew,
-                e->d_ghosts, e->nghost, d_curr, gc, e->d_vdeg,
-                e->d_cinfo, e->d_cupd, e->d_rc_ids, nrc, e->d_rc_info,
-                e->d_rcu, constant, d_target, e->d_cw, e->d_hash_off,
+                e->d_sigma, e->d_sigma_inv, e->d_xadj, e->d_tails,
+                e->d_ghosts, e->nghost, e->d_vcurr, e->d_vghost, e->d_vdeg,
+                e->d_cinfo, e->d_cupd, e->d_rc_ids, e->d_rc_info,
+                e->d_rcu, constant, e->d_vtarget, e->d_cw, e->d_hash_off,
                 e->d_hkeys, e->d_hacc);
         }
         static const bool no_iter1 = getenv("MV_NO_ITER1") != nullptr;
@@ -2537,18 +2600,29 @@ double mv_engine_run(mv_engine *e, double lower, double thresh,
                 dispatch_slots(std::integral_constant<bool, false>{}, s0, s1);
             }
         };
+        // persist the sweep's u32 view targets as labels (p>1 only)
+        auto writeback = [&](i64 s0, i64 s1) {
+            if (p == 1 || s1 <= s0) return;
+            k_view_to_labels<<<grid_for(s1 - s0, 256, 2048), 256, 0, st>>>(
+                s0, s1, e->d_perm, e->d_vtarget, e->d_sigma, e->base,
+                e->d_rc_ids, lnv, d_target);
+        };
         if (p > 1 && e->overlap) {
-            // exports first; the interior launch goes in BEFORE the halo's
+            // exports first (sweep + label writeback, so halo #1a can
+            // gather them); the interior launch goes in BEFORE the halo's
             // host-side work so it runs under the exchange
             sweep_range(0, e->nexp);
+            writeback(0, e->nexp);
             HIP_CHECK(hipEventRecord(e->ev_p1, st));
             sweep_range(e->nexp, lnv);
+            writeback(e->nexp, lnv);
             HIP_CHECK(hipStreamWaitEvent(e->stream2, e->ev_p1, 0));
-            run_halo1a(e, d_target, e->d_gc[gcur ^ 1],
-                       e->comm2 ? e->comm2 : e->comm, e->stream2);
+            run_halo1a(e, d_target, e->comm2 ? e->comm2 : e->comm,
+                       e->stream2);
             HIP_CHECK(hipEventRecord(e->ev_halo, e->stream2));
         } else {
             sweep_range(e->nhi, lnv);
+            writeback(0, lnv);
         }
         HIP_CHECK(hipEventRecord(ev1, st));
         PHASE("sweep-done");
@@ -2644,8 +2718,6 @@ double mv_engine_run(mv_engine *e, double lower, double thresh,
         d_past = d_curr;
         d_curr = d_target;
         d_target = tmp;
-        if (e->overlap) gcur ^= 1; // next iteration reads the #1a buffer
-                                   // filled during this one
         if (numIters >= 10000) break; // safety net, never hit in practice
     }
     // overlap mode posted one speculative #1a past the exit; every rank

@@ -128,6 +128,93 @@ def test_loopback_parity_no_delta_p4():
         del os.environ["MV_NO_DELTA"]
 
 
+def _zipf_graph(nv, max_deg, seed=7, weighted=False):
+    rng = np.random.default_rng(seed)
+    deg = np.minimum((2.0 / rng.power(2.0, nv)).astype(np.int64), max_deg)
+    src = np.repeat(np.arange(nv), deg)
+    dst = rng.integers(0, nv, src.size)
+    keep = src != dst
+    src, dst = src[keep], dst[keep]
+    u = np.concatenate([src, dst])
+    v = np.concatenate([dst, src])
+    if weighted:
+        w0 = rng.uniform(0.01, 1.0, src.size)
+        w = np.concatenate([w0, w0])
+    else:
+        w = None
+    order = np.lexsort((v, u))
+    u, v = u[order], v[order]
+    if w is not None:
+        w = w[order]
+    xadj = np.zeros(nv + 1, dtype=np.int64)
+    np.add.at(xadj, u + 1, 1)
+    xadj = np.cumsum(xadj)
+    return xadj, v, w
+
+
+@pytest.mark.parametrize("weighted", [False, True])
+def test_loopback_skewed_p2(weighted):
+    """Skewed (hub-heavy) graph partitioned 2 ways: exercises the
+    multi-rank wave-per-vertex hub kernel (unit) / the spill path (-w)
+    under the view encoding; engine vs oracle on identical CSRs."""
+    from minivite_amd import Graph, Engine, LoopbackSession
+    from oracle.oracle import OracleGraph, louvain, sha
+    nv, world = 50000, 2
+    xadj, tails, w = _zipf_graph(nv, 5000, weighted=weighted)
+    parts = np.array([(nv * r) // world for r in range(world + 1)],
+                     dtype=np.int64)
+    csrs = []
+    for r in range(world):
+        lo, hi = parts[r], parts[r + 1]
+        xa = (xadj[lo:hi + 1] - xadj[lo]).copy()
+        ta = tails[xadj[lo]:xadj[hi]].copy()
+        wa = None if w is None else w[xadj[lo]:xadj[hi]].copy()
+        csrs.append((xa, ta, wa))
+    og = OracleGraph.from_csr(nv, world, parts, csrs)
+    omod, oiters, ott, otm = louvain(og, trace=True, trace_cap=300)
+    og.free()
+
+    ses = LoopbackSession(world)
+    results = {}
+    errors = []
+
+    def rank_main(r):
+        try:
+            xa, ta, wa = csrs[r]
+            g = Graph.from_csr(nv, r, world, parts, xa, ta, wa)
+            e = Engine.loopback(ses, r, device=0)
+            e.load_graph(g)
+            e.set_trace(300)
+            mod, iters = e.run()
+            tt, _ = e.trace(iters)
+            results[r] = (mod, iters, tt.copy())
+            e.destroy()
+            g.free()
+        except Exception as ex:  # pragma: no cover
+            errors.append((r, repr(ex)))
+
+    import threading
+    threads = [threading.Thread(target=rank_main, args=(r,))
+               for r in range(world)]
+    for t in threads:
+        t.start()
+    for t in threads:
+        t.join(timeout=600)
+    ses.destroy()
+    assert not errors, errors
+    iters = results[0][1]
+    assert iters == oiters
+    if weighted:
+        assert abs(results[0][0] - omod) < 1e-9
+    else:
+        assert float(results[0][0]).hex() == float(omod).hex()
+    for k in range(min(iters, 300)):
+        full = np.zeros(nv, dtype=np.int64)
+        for r in range(world):
+            full[parts[r]:parts[r + 1]] = results[r][2][k]
+        assert sha(full) == sha(ott[k]), f"iteration {k+1}"
+
+
 def test_loopback_deterministic_p4():
     """Two identical p=4 loopback runs agree bit-for-bit (per-sender
     in-order delta application — run-to-run determinism at nranks > 2)."""
