@@ -1792,7 +1792,11 @@ static void rccl_alltoallv(mv_engine *e, const void *send, const i64 *soff,
             const auto &ps = s->posts[r];
             const i64 cnt = ps.soff[e->rank + 1] - ps.soff[e->rank];
             if (cnt != roff[r + 1] - roff[r]) {
-                std::fprintf(stderr, "loopback alltoallv count mismatch\n");
+                std::fprintf(stderr,
+                             "loopback alltoallv count mismatch: me=%d r=%d "
+                             "sender says %lld, I expect %lld (eb=%zu)\n",
+                             e->rank, r, (long long)cnt,
+                             (long long)(roff[r + 1] - roff[r]), elem_bytes);
                 std::abort();
             }
             if (cnt > 0)
@@ -2313,7 +2317,16 @@ double mv_engine_run(mv_engine *e, double lower, double thresh,
             e->ssz = 0;
         }
 
-        if (p > 1) build_sell(e); // p==1: built at load
+        if (p > 1) {
+            build_sell(e); // p==1: built at load
+            // overlap changes the collective sequence, so it must be a
+            // GLOBAL decision: all ranks take it or none (a rank can be
+            // locally skewed / export-free while others are not)
+            std::vector<i64> flag(p, e->overlap ? 1 : 0), fm;
+            exchange_counts(e, flag, fm);
+            for (int r = 0; r < p; r++)
+                if (fm[(size_t)r * p] == 0) e->overlap = 0;
+        }
     }
 
     PHASE("setup-done");
