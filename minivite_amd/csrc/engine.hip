@@ -847,6 +847,161 @@ __global__ __launch_bounds__(256) void k4_sweep_hi_mr(
     }
 }
 
+// ---- K4 hub path for WEIGHTED skewed graphs (per-lane + hash) ----
+// The wave-per-vertex hub kernel needs order-free sums (unit weights); -w
+// hubs instead keep one LANE per vertex — accumulation stays in edge
+// order, bit-exact (dspl.hpp:240-271) — but the spill becomes an
+// open-addressed per-VERTEX hash (d_hash_off regions), turning the linear
+// spill's O(deg^2) probing into O(deg) expected. Works for p==1 too: the
+// u32 slot array IS the view there (base 0, no ghosts), so one kernel
+// serves both.
+__global__ __launch_bounds__(256) void k4_sweep_hubw(
+    i64 nhi, i64 lnv, i64 base, const unsigned *__restrict__ perm,
+    const unsigned *__restrict__ deg_int, const i64 *__restrict__ chunk_off,
+    const int *__restrict__ sell_tidx, const double *__restrict__ sell_w,
+    const unsigned *__restrict__ vcurr, const unsigned *__restrict__ vghost,
+    const double *__restrict__ vDegree, const unsigned *__restrict__ sigma,
+    const Cinfo *__restrict__ cinfo, Cinfo *__restrict__ cupd,
+    const i64 *__restrict__ rc_ids, const Info16 *__restrict__ rc_info,
+    Info16 *__restrict__ rcu, double constant,
+    unsigned *__restrict__ vtarget, double *__restrict__ clusterWeight,
+    const i64 *__restrict__ hash_off, i64 *__restrict__ hkeys,
+    double *__restrict__ hacc) {
+    const i64 gthread = blockIdx.x * (i64)blockDim.x + threadIdx.x;
+    const i64 stride = (i64)gridDim.x * blockDim.x;
+    auto label_of = [&](unsigned v) -> i64 {
+        return (v < lnv) ? base + (i64)sigma[v] : rc_ids[clamp0(v - lnv)];
+    };
+    for (i64 s = gthread; s < nhi; s += stride) {
+        const i64 i = perm[s];
+        const int deg = (int)deg_int[i];
+        const i64 ebase = chunk_off[s >> 6] + (s & 63);
+        const unsigned cc = vcurr[i];
+        const i64 hoff = hash_off[s];
+        const i64 cap = hash_off[s + 1] - hoff; // power of two
+        double ccDeg;
+        i64 ccSize;
+        if (cc < lnv) {
+            const Cinfo c = cinfo[cc];
+            ccDeg = c.degree;
+            ccSize = c.size;
+        } else {
+            const Info16 c = rc_info[cc - lnv];
+            ccDeg = c.degree;
+            ccSize = c.size;
+        }
+        double c0 = 0.0, selfLoop = 0.0;
+        int ncand = 0;
+        constexpr int CH = 8;
+        for (int k0 = 0; k0 < deg; k0 += CH) {
+            const int m = min(CH, deg - k0);
+            i64 tb[CH];
+            unsigned cb[CH];
+            double wb[CH];
+#pragma unroll
+            for (int j = 0; j < CH; j++) {
+                const i64 slot = (j < m) ? ebase + (i64)(k0 + j) * 64 : ebase;
+                tb[j] = sell_tidx[slot];
+                wb[j] = sell_w[slot];
+            }
+#pragma unroll
+            for (int j = 0; j < CH; j++)
+                cb[j] = (tb[j] < lnv) ? vcurr[tb[j]]
+                                      : vghost[clamp0(tb[j] - lnv)];
+            for (int j = 0; j < m; j++) {
+                const double w = wb[j];
+                if (tb[j] == i) selfLoop += w; // dspl.hpp:247-248
+                const unsigned tcomm = cb[j];
+                if (tcomm == cc) { c0 += w; continue; }
+                i64 pos = (i64)(((uint64_t)tcomm * 0x9E3779B97F4A7C15ull)
+                                >> 32) & (cap - 1);
+                for (;;) {
+                    const i64 key = hkeys[hoff + pos];
+                    if (key == (i64)tcomm) {
+                        hacc[hoff + pos] += w; // edge-order: one lane owns
+                        break;                 // this vertex
+                    }
+                    if (key == -1) {
+                        hkeys[hoff + pos] = tcomm;
+                        hacc[hoff + pos] = w;
+                        ncand++;
+                        break;
+                    }
+                    pos = (pos + 1) & (cap - 1);
+                }
+            }
+        }
+        clusterWeight[i] = c0; // dspl.hpp:318
+
+        // distGetMaxIndex over the hash (dspl.hpp:174-228). The scan order
+        // is hash-slot order, which is fine: the argmax with the lazy
+        // label tie-break is a total order, order-independent.
+        const double vdeg = vDegree[i];
+        const double eix = c0 - selfLoop;
+        const double ax = ccDeg - vdeg;
+        double maxGain = 0.0;
+        unsigned maxIndex = cc;
+        i64 maxSize = ccSize;
+        int seen = 0;
+        for (i64 t = 0; t < cap && seen < ncand; t++) {
+            const i64 yk = hkeys[hoff + t];
+            if (yk == -1) continue;
+            seen++;
+            const unsigned y = (unsigned)yk;
+            const double eiy = hacc[hoff + t];
+            double ay;
+            i64 ysz;
+            if (y < lnv) {
+                const Cinfo c = cinfo[y];
+                ay = c.degree;
+                ysz = c.size;
+            } else {
+                const Info16 c = rc_info[y - lnv];
+                ay = c.degree;
+                ysz = c.size;
+            }
+            const double curGain =
+                2.0 * (eiy - eix) - 2.0 * vdeg * (ay - ax) * constant;
+            if (curGain > maxGain) {
+                maxGain = curGain;
+                maxIndex = y;
+                maxSize = ysz;
+            } else if (curGain == maxGain && curGain != 0.0) {
+                // real branch, not a select (DESIGN.md §4 miscompile note)
+                if (label_of(y) < label_of(maxIndex)) {
+                    maxIndex = y;
+                    maxSize = ysz;
+                }
+            }
+        }
+        if (maxSize == 1 && ccSize == 1 && maxIndex != cc) { // :224-225
+            if (label_of(maxIndex) > label_of(cc)) maxIndex = cc;
+        }
+        if (deg == 0) maxIndex = cc;
+        if (maxIndex != cc) {
+            if (cc < lnv) {
+                Cinfo *u = &cupd[cc];
+                atomicAdd(&u->degree, -vdeg);
+                atomic_add_i64(&u->size, -1);
+            } else {
+                Info16 *u = &rcu[cc - lnv];
+                atomicAdd(&u->degree, -vdeg);
+                atomic_add_i64(&u->size, -1);
+            }
+            if (maxIndex < lnv) {
+                Cinfo *u = &cupd[maxIndex];
+                atomicAdd(&u->degree, vdeg);
+                atomic_add_i64(&u->size, 1);
+            } else {
+                Info16 *u = &rcu[maxIndex - lnv];
+                atomicAdd(&u->degree, vdeg);
+                atomic_add_i64(&u->size, 1);
+            }
+        }
+        vtarget[i] = maxIndex;
+    }
+}
+
 // ---- K4 first-iteration specialization ----
 // At iteration 1 currComm is the identity (dspl.hpp:145-147), so per vertex:
 // counter[0] collects exactly the self-loop weight (eix = counter[0] -
@@ -1904,8 +2059,8 @@ static void build_sell(mv_engine *e) {
         (void)hipcub::DeviceRadixSort::SortPairsDescending(
             d_tmp, tb, e->d_deg, d_degs, e->d_iota, e->d_perm, lnv, 0, 32,
             st);
-        // high-degree split + per-vertex hash regions (unit weights
-        // only: -w needs the serial edge-order path for bit parity)
+        // high-degree split + per-vertex hash regions (unit: wave-per-
+        // vertex atomic hash; -w: per-lane serial hash in edge order)
         std::vector<unsigned> sdeg(lnv);
         HIP_CHECK(hipMemcpyAsync(sdeg.data(), d_degs, 4 * lnv,
                                  hipMemcpyDeviceToHost, st));
@@ -1913,7 +2068,7 @@ static void build_sell(mv_engine *e) {
         constexpr unsigned HI_THRESH = 256;
         i64 nhi = 0;
         while (nhi < lnv && sdeg[nhi] > HI_THRESH) nhi++;
-        e->nhi = (e->unit_weights && e->skewed) ? nhi : 0;
+        e->nhi = e->skewed ? nhi : 0;
         if (e->nhi > 0) {
             std::vector<i64> hoff(e->nhi + 1);
             i64 acc = 0;
@@ -2578,10 +2733,19 @@ double mv_engine_run(mv_engine *e, double lower, double thresh,
                     e->d_ghosts, e->d_vdeg, e->d_sigma, e->d_cupd,
                     e->d_rc_info, e->d_rcu, constant, e->d_vtarget, e->d_cw);
         };
-        if (e->nhi > 0) { // wave-per-vertex hubs (unit weights only)
+        if (e->nhi > 0) { // hub path (skewed graphs)
             HIP_CHECK(hipMemsetAsync(e->d_hkeys, 0xFF, 8 * e->hash_total, st));
             HIP_CHECK(hipMemsetAsync(e->d_hacc, 0, 8 * e->hash_total, st));
-            if (p == 1)
+            if (!e->unit_weights) // -w: per-lane serial hash (edge order)
+                k4_sweep_hubw<<<grid_for(e->nhi), 256, 0, st>>>(
+                    e->nhi, lnv, e->base, e->d_perm, e->d_deg,
+                    e->d_chunk_off, e->d_sell_tidx, e->d_sell_w,
+                    p == 1 ? (const unsigned *)d_curr : e->d_vcurr,
+                    e->d_vghost, e->d_vdeg, e->d_sigma, e->d_cinfo,
+                    e->d_cupd, e->d_rc_ids, e->d_rc_info, e->d_rcu,
+                    constant, p == 1 ? (unsigned *)d_target : e->d_vtarget,
+                    e->d_cw, e->d_hash_off, e->d_hkeys, e->d_hacc);
+            else if (p == 1)
                 k4_sweep_hi_p1<<<grid_for(e->nhi * 64, 256, 2048), 256, 0,
                                  st>>>(
                     e->nhi, lnv, e->d_perm, e->d_deg, e->d_sigma,

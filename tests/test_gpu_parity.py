@@ -203,6 +203,49 @@ def test_skewed_degree_graph_parity():
     print("skewed: iters", iters, "sweep_ms", st["sweep_ms"])
 
 
+def test_skewed_weighted_graph_parity():
+    """Weighted power-law graph (hub degree ~10k): exercises the per-lane
+    hash-spill hub path (k4_sweep_hubw — bit-exact edge-order -w sums with
+    O(deg) probing); engine vs oracle on the identical from_csr input."""
+    import numpy as np
+    from minivite_amd import Graph, Engine
+    from oracle.oracle import OracleGraph, louvain, sha
+    rng = np.random.default_rng(11)
+    nv = 100000
+    deg = np.minimum((2.0 / rng.power(2.0, nv)).astype(np.int64), 10000)
+    src = np.repeat(np.arange(nv), deg)
+    dst = rng.integers(0, nv, src.size)
+    keep = src != dst
+    src, dst = src[keep], dst[keep]
+    w0 = rng.uniform(0.01, 1.0, src.size)
+    u = np.concatenate([src, dst])
+    v = np.concatenate([dst, src])
+    w = np.concatenate([w0, w0])
+    order = np.lexsort((v, u))
+    u, v, w = u[order], v[order], w[order]
+    xadj = np.zeros(nv + 1, dtype=np.int64)
+    np.add.at(xadj, u + 1, 1)
+    xadj = np.cumsum(xadj)
+    parts = np.array([0, nv], dtype=np.int64)
+    og = OracleGraph.from_csr(nv, 1, parts, [(xadj, v, w)])
+    omod, oiters, ott, otm = louvain(og, trace=True, trace_cap=300)
+    og.free()
+    g = Graph.from_csr(nv, 0, 1, parts, xadj, v, w)
+    e = Engine(device=0)
+    e.load_graph(g)
+    e.set_trace(300)
+    mod, iters = e.run()
+    tt, tm = e.trace(iters)
+    st = e.stats()
+    e.destroy()
+    g.free()
+    assert iters == oiters
+    assert abs(mod - omod) < 1e-9
+    for k in range(min(iters, 300)):
+        assert sha(tt[k]) == sha(ott[k]), f"iteration {k+1}"
+    print("skewed-w: iters", iters, "sweep_ms", st["sweep_ms"])
+
+
 def test_binary_file_path_parity():
     """config-5 style path: write .bin, read back partitioned (-b balanced),
     run the engine on the read graph — must match the in-memory graph's
