@@ -236,6 +236,199 @@ __global__ __launch_bounds__(256) void v3_probe(
     }
 }
 
+// ---------------- V0u: u32-slot replica of the CURRENT production
+// kernel (round-2 slot/view mode: 4-B community gathers, 12 B/lane LDS)
+__global__ __launch_bounds__(256) void v0u(
+    i64 lnv, const unsigned *__restrict__ deg, const i64 *__restrict__ coff,
+    const int *__restrict__ sell, const unsigned *__restrict__ comm,
+    const double *__restrict__ aux, i64 *__restrict__ out) {
+    extern __shared__ char smem[];
+    double *sacc = (double *)smem;
+    unsigned *skey = (unsigned *)(smem + 8 * SLOTS * blockDim.x);
+    const int tid = threadIdx.x;
+    const i64 g0 = blockIdx.x * (i64)blockDim.x + threadIdx.x;
+    const i64 stride = (i64)gridDim.x * blockDim.x;
+    for (i64 s = g0; s < lnv; s += stride) {
+        const int d = (int)deg[s];
+        const i64 eb = coff[s >> 6] + (s & 63);
+        const unsigned cc = comm[s];
+        double c0 = 0.0;
+        int ns = 0;
+        for (int k0 = 0; k0 < d; k0 += CH) {
+            const int m = min(CH, d - k0);
+            i64 tb[CH];
+            unsigned cb[CH];
+#pragma unroll
+            for (int j = 0; j < CH; j++)
+                tb[j] = sell[(j < m) ? eb + (i64)(k0 + j) * 64 : eb];
+#pragma unroll
+            for (int j = 0; j < CH; j++) cb[j] = comm[tb[j]];
+            for (int j = 0; j < m; j++) {
+                const unsigned tc = cb[j];
+                if (tc == cc) { c0 += 1.0; continue; }
+                bool f = false;
+                for (int t = 0; t < ns; t++)
+                    if (skey[t * blockDim.x + tid] == tc) {
+                        sacc[t * blockDim.x + tid] += 1.0;
+                        f = true;
+                        break;
+                    }
+                if (!f && ns < SLOTS) {
+                    skey[ns * blockDim.x + tid] = tc;
+                    sacc[ns * blockDim.x + tid] = 1.0;
+                    ns++;
+                }
+            }
+        }
+        double bg = 0.0;
+        unsigned bl = cc;
+        for (int t = 0; t < ns; t++) {
+            const unsigned y = skey[t * blockDim.x + tid];
+            const double eiy = sacc[t * blockDim.x + tid];
+            const double g = 2.0 * eiy - 1e-7 * aux[y & (lnv - 1)];
+            if (g > bg) { bg = g; bl = y; }
+        }
+        out[s] = (i64)bl + (i64)c0;
+    }
+}
+
+// ---------------- V4: first RS candidate slots in REGISTERS ----------------
+// The probe loop's LDS round-trips per edge are the measured cost center;
+// most steady-state vertices carry <= 4 distinct candidates, so keep those
+// in VGPRs (fully unrolled compares) and overflow into LDS.
+template <int RS, int LS>
+__global__ __launch_bounds__(256) void v4(
+    i64 lnv, const unsigned *__restrict__ deg, const i64 *__restrict__ coff,
+    const int *__restrict__ sell, const unsigned *__restrict__ comm,
+    const double *__restrict__ aux, i64 *__restrict__ out) {
+    extern __shared__ char smem[];
+    double *sacc = (double *)smem;
+    unsigned *skey = (unsigned *)(smem + 8 * LS * blockDim.x);
+    const int tid = threadIdx.x;
+    const i64 g0 = blockIdx.x * (i64)blockDim.x + threadIdx.x;
+    const i64 stride = (i64)gridDim.x * blockDim.x;
+    for (i64 s = g0; s < lnv; s += stride) {
+        const int d = (int)deg[s];
+        const i64 eb = coff[s >> 6] + (s & 63);
+        const unsigned cc = comm[s];
+        double c0 = 0.0;
+        unsigned rkey[RS];
+        double racc[RS];
+        int ns = 0; // total candidates (registers first, then LDS)
+        for (int k0 = 0; k0 < d; k0 += CH) {
+            const int m = min(CH, d - k0);
+            i64 tb[CH];
+            unsigned cb[CH];
+#pragma unroll
+            for (int j = 0; j < CH; j++)
+                tb[j] = sell[(j < m) ? eb + (i64)(k0 + j) * 64 : eb];
+#pragma unroll
+            for (int j = 0; j < CH; j++) cb[j] = comm[tb[j]];
+            for (int j = 0; j < m; j++) {
+                const unsigned tc = cb[j];
+                if (tc == cc) { c0 += 1.0; continue; }
+                bool f = false;
+#pragma unroll
+                for (int t = 0; t < RS; t++)
+                    if (t < ns && rkey[t] == tc) {
+                        racc[t] += 1.0;
+                        f = true;
+                    }
+                if (f) continue;
+                for (int t = RS; t < ns; t++)
+                    if (skey[(t - RS) * blockDim.x + tid] == tc) {
+                        sacc[(t - RS) * blockDim.x + tid] += 1.0;
+                        f = true;
+                        break;
+                    }
+                if (f) continue;
+                if (ns < RS) {
+                    rkey[ns] = tc;
+                    racc[ns] = 1.0;
+                    ns++;
+                } else if (ns < RS + LS) {
+                    skey[(ns - RS) * blockDim.x + tid] = tc;
+                    sacc[(ns - RS) * blockDim.x + tid] = 1.0;
+                    ns++;
+                }
+            }
+        }
+        double bg = 0.0;
+        unsigned bl = cc;
+#pragma unroll
+        for (int t = 0; t < RS; t++) {
+            if (t >= ns) break;
+            const double g = 2.0 * racc[t] - 1e-7 * aux[rkey[t] & (lnv - 1)];
+            if (g > bg) { bg = g; bl = rkey[t]; }
+        }
+        for (int t = RS; t < ns; t++) {
+            const unsigned y = skey[(t - RS) * blockDim.x + tid];
+            const double g = 2.0 * sacc[(t - RS) * blockDim.x + tid] -
+                             1e-7 * aux[y & (lnv - 1)];
+            if (g > bg) { bg = g; bl = y; }
+        }
+        out[s] = (i64)bl + (i64)c0;
+    }
+}
+
+// ---------------- V7: 32-bit mode probe (USE_32_BIT_GRAPH analog) ------
+// Same structure as V0u but float accumulators + float aux gathers: the
+// arithmetic/data shape the reference's 32-bit build would give. Measures
+// whether narrower floats pay on this latency-bound kernel.
+__global__ __launch_bounds__(256) void v7(
+    i64 lnv, const unsigned *__restrict__ deg, const i64 *__restrict__ coff,
+    const int *__restrict__ sell, const unsigned *__restrict__ comm,
+    const float *__restrict__ aux32, i64 *__restrict__ out) {
+    extern __shared__ char smem[];
+    float *sacc = (float *)smem;
+    unsigned *skey = (unsigned *)(smem + 4 * SLOTS * blockDim.x);
+    const int tid = threadIdx.x;
+    const i64 g0 = blockIdx.x * (i64)blockDim.x + threadIdx.x;
+    const i64 stride = (i64)gridDim.x * blockDim.x;
+    for (i64 s = g0; s < lnv; s += stride) {
+        const int d = (int)deg[s];
+        const i64 eb = coff[s >> 6] + (s & 63);
+        const unsigned cc = comm[s];
+        float c0 = 0.0f;
+        int ns = 0;
+        for (int k0 = 0; k0 < d; k0 += CH) {
+            const int m = min(CH, d - k0);
+            i64 tb[CH];
+            unsigned cb[CH];
+#pragma unroll
+            for (int j = 0; j < CH; j++)
+                tb[j] = sell[(j < m) ? eb + (i64)(k0 + j) * 64 : eb];
+#pragma unroll
+            for (int j = 0; j < CH; j++) cb[j] = comm[tb[j]];
+            for (int j = 0; j < m; j++) {
+                const unsigned tc = cb[j];
+                if (tc == cc) { c0 += 1.0f; continue; }
+                bool f = false;
+                for (int t = 0; t < ns; t++)
+                    if (skey[t * blockDim.x + tid] == tc) {
+                        sacc[t * blockDim.x + tid] += 1.0f;
+                        f = true;
+                        break;
+                    }
+                if (!f && ns < SLOTS) {
+                    skey[ns * blockDim.x + tid] = tc;
+                    sacc[ns * blockDim.x + tid] = 1.0f;
+                    ns++;
+                }
+            }
+        }
+        float bg = 0.0f;
+        unsigned bl = cc;
+        for (int t = 0; t < ns; t++) {
+            const unsigned y = skey[t * blockDim.x + tid];
+            const float eiy = sacc[t * blockDim.x + tid];
+            const float g = 2.0f * eiy - 1e-7f * aux32[y & (lnv - 1)];
+            if (g > bg) { bg = g; bl = y; }
+        }
+        out[s] = (i64)bl + (i64)c0;
+    }
+}
+
 int main(int argc, char **argv) {
     const i64 lnv = argc > 1 ? atoll(argv[1]) : (1ll << 24);
     const i64 nchunks = (lnv + 63) / 64;
@@ -324,6 +517,45 @@ int main(int argc, char **argv) {
         v3_gather<<<2048, 256>>>(elems, d_sell, d_comm, d_tcomm);
         v3_probe<<<grid, 256, SLOTS * 256 * 16>>>(lnv, d_deg, d_coff, d_tcomm,
                                                   d_comm, d_aux, d_out);
+    });
+
+    // u32 community image for the slot/view-mode variants
+    unsigned *d_comm32;
+    HC(hipMalloc(&d_comm32, 4 * lnv));
+    {
+        std::vector<unsigned> c32(lnv);
+        for (i64 i = 0; i < lnv; i++) c32[i] = (unsigned)comm[i];
+        HC(hipMemcpy(d_comm32, c32.data(), 4 * lnv, hipMemcpyHostToDevice));
+    }
+    bench("V0u-u32", [&] {
+        v0u<<<grid, 256, SLOTS * 256 * 12>>>(lnv, d_deg, d_coff, d_sell,
+                                             d_comm32, d_aux, d_out);
+    });
+    bench("V4-r4l8", [&] {
+        v4<4, 8><<<grid, 256, 8 * 256 * 12>>>(lnv, d_deg, d_coff, d_sell,
+                                              d_comm32, d_aux, d_out);
+    });
+    bench("V4-r4l4", [&] {
+        v4<4, 4><<<grid, 256, 4 * 256 * 12>>>(lnv, d_deg, d_coff, d_sell,
+                                              d_comm32, d_aux, d_out);
+    });
+    bench("V4-r6l4", [&] {
+        v4<6, 4><<<grid, 256, 4 * 256 * 12>>>(lnv, d_deg, d_coff, d_sell,
+                                              d_comm32, d_aux, d_out);
+    });
+    bench("V4-r8l4", [&] {
+        v4<8, 4><<<grid, 256, 4 * 256 * 12>>>(lnv, d_deg, d_coff, d_sell,
+                                              d_comm32, d_aux, d_out);
+    });
+    float *d_aux32;
+    HC(hipMalloc(&d_aux32, 4 * lnv));
+    {
+        std::vector<float> a32(lnv, 1.0f);
+        HC(hipMemcpy(d_aux32, a32.data(), 4 * lnv, hipMemcpyHostToDevice));
+    }
+    bench("V7-f32", [&] {
+        v7<<<grid, 256, SLOTS * 256 * 8>>>(lnv, d_deg, d_coff, d_sell,
+                                           d_comm32, d_aux32, d_out);
     });
     return 0;
 }
