@@ -1857,8 +1857,14 @@ int mv_engine_load_graph(mv_engine *e, const mv_graph *g) {
     HIP_CHECK(hipMalloc(&e->d_tails, 8 * std::max<i64>(lne, 1)));
     HIP_CHECK(hipMemcpy(e->d_tails, mv_graph_tails(g), 8 * lne,
                         hipMemcpyHostToDevice));
-    HIP_CHECK(hipMalloc(&e->d_ew, 8 * std::max<i64>(lne, 1)));
-    HIP_CHECK(hipMemcpy(e->d_ew, w, 8 * lne, hipMemcpyHostToDevice));
+    // unit graphs never read edge weights on device (K1 counts rows, the
+    // SELL carries no weight stream): skip the 8*lne upload entirely
+    if (e->unit_weights) {
+        HIP_CHECK(hipMalloc(&e->d_ew, 16)); // never-null dummy
+    } else {
+        HIP_CHECK(hipMalloc(&e->d_ew, 8 * std::max<i64>(lne, 1)));
+        HIP_CHECK(hipMemcpy(e->d_ew, w, 8 * lne, hipMemcpyHostToDevice));
+    }
 
     // internal layout: the builder's spatial hint, or identity (the
     // degree-sort fallback then runs in the per-run setup)
