@@ -429,6 +429,73 @@ __global__ __launch_bounds__(256) void v7(
     }
 }
 
+// ---------------- V8: two-phase gain scan ----------------
+// Identical probe to V0u, but the gain scan prefetches ALL candidate-info
+// gathers into registers first (batched independent loads), then runs the
+// compare loop — forcing memory-level parallelism the serial scan leaves
+// to the scheduler.
+__global__ __launch_bounds__(256) void v8(
+    i64 lnv, const unsigned *__restrict__ deg, const i64 *__restrict__ coff,
+    const int *__restrict__ sell, const unsigned *__restrict__ comm,
+    const double *__restrict__ aux, i64 *__restrict__ out) {
+    extern __shared__ char smem[];
+    double *sacc = (double *)smem;
+    unsigned *skey = (unsigned *)(smem + 8 * SLOTS * blockDim.x);
+    const int tid = threadIdx.x;
+    const i64 g0 = blockIdx.x * (i64)blockDim.x + threadIdx.x;
+    const i64 stride = (i64)gridDim.x * blockDim.x;
+    for (i64 s = g0; s < lnv; s += stride) {
+        const int d = (int)deg[s];
+        const i64 eb = coff[s >> 6] + (s & 63);
+        const unsigned cc = comm[s];
+        double c0 = 0.0;
+        int ns = 0;
+        for (int k0 = 0; k0 < d; k0 += CH) {
+            const int m = min(CH, d - k0);
+            i64 tb[CH];
+            unsigned cb[CH];
+#pragma unroll
+            for (int j = 0; j < CH; j++)
+                tb[j] = sell[(j < m) ? eb + (i64)(k0 + j) * 64 : eb];
+#pragma unroll
+            for (int j = 0; j < CH; j++) cb[j] = comm[tb[j]];
+            for (int j = 0; j < m; j++) {
+                const unsigned tc = cb[j];
+                if (tc == cc) { c0 += 1.0; continue; }
+                bool f = false;
+                for (int t = 0; t < ns; t++)
+                    if (skey[t * blockDim.x + tid] == tc) {
+                        sacc[t * blockDim.x + tid] += 1.0;
+                        f = true;
+                        break;
+                    }
+                if (!f && ns < SLOTS) {
+                    skey[ns * blockDim.x + tid] = tc;
+                    sacc[ns * blockDim.x + tid] = 1.0;
+                    ns++;
+                }
+            }
+        }
+        double bg = 0.0;
+        unsigned bl = cc;
+        double av[SLOTS];
+        unsigned ky[SLOTS];
+        double ac[SLOTS];
+#pragma unroll
+        for (int t = 0; t < SLOTS; t++) {
+            const int tt = t < ns ? t : 0;
+            ky[t] = skey[tt * blockDim.x + tid];
+            ac[t] = sacc[tt * blockDim.x + tid];
+            av[t] = aux[ky[t] & (lnv - 1)];
+        }
+        for (int t = 0; t < ns; t++) {
+            const double g = 2.0 * ac[t] - 1e-7 * av[t];
+            if (g > bg) { bg = g; bl = ky[t]; }
+        }
+        out[s] = (i64)bl + (i64)c0;
+    }
+}
+
 int main(int argc, char **argv) {
     const i64 lnv = argc > 1 ? atoll(argv[1]) : (1ll << 24);
     const i64 nchunks = (lnv + 63) / 64;
@@ -553,6 +620,10 @@ int main(int argc, char **argv) {
         std::vector<float> a32(lnv, 1.0f);
         HC(hipMemcpy(d_aux32, a32.data(), 4 * lnv, hipMemcpyHostToDevice));
     }
+    bench("V8-2ph", [&] {
+        v8<<<grid, 256, SLOTS * 256 * 12>>>(lnv, d_deg, d_coff, d_sell,
+                                            d_comm32, d_aux, d_out);
+    });
     bench("V7-f32", [&] {
         v7<<<grid, 256, SLOTS * 256 * 8>>>(lnv, d_deg, d_coff, d_sell,
                                            d_comm32, d_aux32, d_out);
