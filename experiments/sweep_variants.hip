@@ -17,6 +17,7 @@
 // Build: hipcc --offload-arch=gfx950 -O3 -std=c++17 sweep_variants.hip -o sweep_variants
 // Run:   ./sweep_variants [nv]
 
+#include <algorithm>
 #include <chrono>
 #include <cstdint>
 #include <cstdio>
@@ -628,5 +629,29 @@ int main(int argc, char **argv) {
         v7<<<grid, 256, SLOTS * 256 * 8>>>(lnv, d_deg, d_coff, d_sell,
                                            d_comm32, d_aux32, d_out);
     });
+
+    // V9: V0u over per-row INTERNAL-sorted tails (ascending gather
+    // addresses within the spatial window): measures pure locality.
+    {
+        std::vector<int> sell2(sell);
+        std::vector<int> row;
+        for (i64 s = 0; s < lnv; s++) {
+            const i64 eb = coff[s >> 6] + (s & 63);
+            row.clear();
+            for (unsigned k = 0; k < deg[s]; k++)
+                row.push_back(sell2[eb + (i64)k * 64]);
+            std::sort(row.begin(), row.end());
+            for (unsigned k = 0; k < deg[s]; k++)
+                sell2[eb + (i64)k * 64] = row[k];
+        }
+        int *d_sell2;
+        HC(hipMalloc(&d_sell2, 4 * elems));
+        HC(hipMemcpy(d_sell2, sell2.data(), 4 * elems,
+                     hipMemcpyHostToDevice));
+        bench("V9-sorted", [&] {
+            v0u<<<grid, 256, SLOTS * 256 * 12>>>(lnv, d_deg, d_coff, d_sell2,
+                                                 d_comm32, d_aux, d_out);
+        });
+    }
     return 0;
 }
