@@ -215,6 +215,68 @@ def test_loopback_skewed_p2(weighted):
         assert sha(full) == sha(ott[k]), f"iteration {k+1}"
 
 
+def test_loopback_balanced_read_p4():
+    """-b edge-balanced reader (find_balanced_num_edges partition,
+    graph.hpp:416-461) feeding the ENGINE at p=4 on one GPU: engine vs
+    oracle on the identical balanced partition of the same .bin."""
+    import tempfile
+    from minivite_amd import Graph, Engine, LoopbackSession, lib
+    from oracle.oracle import OracleGraph, louvain, sha
+    nv, world = 16384, 4
+    g0 = Graph.rgg(nv, 0, 1)
+    with tempfile.TemporaryDirectory() as d:
+        path = os.path.join(d, "g.bin")
+        g0.write_binary(path)
+        g0.free()
+        graphs, parts = [], None
+        for r in range(world):
+            gr = Graph.read_binary(path, r, world, balanced=True)
+            if parts is None:
+                pp = lib().mv_graph_parts(gr.h)
+                parts = np.array([pp[k] for k in range(world + 1)],
+                                 dtype=np.int64)
+            graphs.append(gr)
+        og = OracleGraph.from_csr(nv, world, parts,
+                                  [g.arrays() for g in graphs])
+        omod, oiters, ott, otm = louvain(og, trace=True)
+        og.free()
+
+        ses = LoopbackSession(world)
+        results = {}
+        errors = []
+
+        def rank_main(r):
+            try:
+                e = Engine.loopback(ses, r, device=0)
+                e.load_graph(graphs[r])
+                e.set_trace(64)
+                mod, iters = e.run()
+                tt, _ = e.trace(iters)
+                results[r] = (mod, iters, tt.copy())
+                e.destroy()
+            except Exception as ex:  # pragma: no cover
+                errors.append((r, repr(ex)))
+
+        threads = [threading.Thread(target=rank_main, args=(r,))
+                   for r in range(world)]
+        for t in threads:
+            t.start()
+        for t in threads:
+            t.join(timeout=600)
+        ses.destroy()
+        for g in graphs:
+            g.free()
+    assert not errors, errors
+    iters = results[0][1]
+    assert iters == oiters
+    assert float(results[0][0]).hex() == float(omod).hex()
+    for k in range(min(iters, 64)):
+        full = np.zeros(nv, dtype=np.int64)
+        for r in range(world):
+            full[parts[r]:parts[r + 1]] = results[r][2][k]
+        assert sha(full) == sha(ott[k]), f"iteration {k+1}"
+
+
 def test_loopback_deterministic_p4():
     """Two identical p=4 loopback runs agree bit-for-bit (per-sender
     in-order delta application — run-to-run determinism at nranks > 2)."""

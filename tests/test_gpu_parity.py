@@ -247,10 +247,12 @@ def test_skewed_weighted_graph_parity():
 
 
 def test_binary_file_path_parity():
-    """config-5 style path: write .bin, read back partitioned (-b balanced),
+    """config-5 style path: write .bin, read back whole (unbalanced p=1),
     run the engine on the read graph — must match the in-memory graph's
     pinned result (exercises mv_graph_read_binary + a hint-less engine
-    load, i.e. the degree-sorted internal order)."""
+    load, i.e. the degree-sorted internal order). The -b balanced path
+    under the engine is covered at p=4 by
+    test_gpu_loopback.test_loopback_balanced_read_p4."""
     import json
     import tempfile
     from minivite_amd import Graph, Engine
