@@ -278,6 +278,29 @@ __global__ void k_fill_sell(i64 lnv, const unsigned *__restrict__ perm,
     }
 }
 
+__global__ void k_scatter_mark(i64 n, const unsigned *__restrict__ idx,
+                               unsigned char *__restrict__ mark) {
+    for (i64 k = blockIdx.x * (i64)blockDim.x + threadIdx.x; k < n;
+         k += (i64)gridDim.x * blockDim.x)
+        mark[idx[k]] = 1;
+}
+
+__global__ void k_gather_flags(i64 lnv, const unsigned *__restrict__ perm,
+                               const unsigned char *__restrict__ mark,
+                               unsigned char *__restrict__ flags,
+                               int invert) {
+    for (i64 s = blockIdx.x * (i64)blockDim.x + threadIdx.x; s < lnv;
+         s += (i64)gridDim.x * blockDim.x)
+        flags[s] = mark[perm[s]] ^ invert;
+}
+
+__global__ void k_spill_uniform(i64 nthreads, i64 per,
+                                i64 *__restrict__ off) {
+    for (i64 t = blockIdx.x * (i64)blockDim.x + threadIdx.x; t < nthreads;
+         t += (i64)gridDim.x * blockDim.x)
+        off[t] = t * per;
+}
+
 // Per-row insertion sort of the SELL image by INTERNAL tail index.
 // UNIT graphs only: their per-community sums are integer-exact under any
 // accumulation order and the argmax tie-break is a total order on labels,
@@ -1606,6 +1629,7 @@ struct mv_engine {
 
     // graph (device)
     i64 nv = 0, lnv = 0, lne = 0, base = 0, bound = 0;
+    int sort_bits = 64; // ceil(log2(nv)): radix sort width for id keys
     std::vector<i64> parts_h;
     i64 *d_parts = nullptr;
     i64 *d_xadj = nullptr;
@@ -1880,6 +1904,8 @@ int mv_engine_load_graph(mv_engine *e, const mv_graph *g) {
     }
     e->lnv = mv_graph_lnv(g);
     e->lne = mv_graph_lne(g);
+    e->sort_bits = 1;
+    while ((1ll << e->sort_bits) < e->nv) e->sort_bits++;
     e->parts_h.assign(mv_graph_parts(g), mv_graph_parts(g) + e->nranks + 1);
     e->base = e->parts_h[e->rank];
     e->bound = e->parts_h[e->rank + 1];
@@ -2164,25 +2190,42 @@ static void build_sell(mv_engine *e) {
     e->overlap = 0;
     if (e->nranks > 1 && !e->skewed && e->ssz > 0 && (e->comm2 || e->lb) &&
         !getenv("MV_NO_OVERLAP")) {
-        std::vector<unsigned> perm_h(lnv);
-        HIP_CHECK(hipMemcpyAsync(perm_h.data(), e->d_perm, 4 * lnv,
-                                 hipMemcpyDeviceToHost, st));
-        std::vector<unsigned> sv(e->ssz);
-        HIP_CHECK(hipMemcpyAsync(sv.data(), e->d_svdata_int, 4 * e->ssz,
-                                 hipMemcpyDeviceToHost, st));
+        // device-side stable partition (two hipCUB Flagged selects): the
+        // host round trip cost ~ms per run at 8-GPU sizes
+        unsigned char *d_mark = nullptr, *d_flags = nullptr;
+        unsigned *d_out = nullptr;
+        i64 *d_num = nullptr;
+        HIP_CHECK(hipMalloc(&d_mark, lnv));
+        HIP_CHECK(hipMalloc(&d_flags, lnv));
+        HIP_CHECK(hipMalloc(&d_out, 4 * lnv));
+        HIP_CHECK(hipMalloc(&d_num, 8));
+        HIP_CHECK(hipMemsetAsync(d_mark, 0, lnv, st));
+        k_scatter_mark<<<grid_for(e->ssz), 256, 0, st>>>(
+            e->ssz, e->d_svdata_int, d_mark);
+        k_gather_flags<<<grid_for(lnv), 256, 0, st>>>(lnv, e->d_perm, d_mark,
+                                                      d_flags, 0);
+        size_t tsel = 0;
+        (void)hipcub::DeviceSelect::Flagged(nullptr, tsel, e->d_perm,
+                                            d_flags, d_out, d_num, lnv, st);
+        void *d_tsel = nullptr;
+        HIP_CHECK(hipMalloc(&d_tsel, std::max<size_t>(tsel, 1)));
+        (void)hipcub::DeviceSelect::Flagged(d_tsel, tsel, e->d_perm, d_flags,
+                                            d_out, d_num, lnv, st);
+        HIP_CHECK(hipMemcpyAsync(&e->nexp, d_num, 8, hipMemcpyDeviceToHost,
+                                 st));
         HIP_CHECK(hipStreamSynchronize(st));
-        std::vector<char> mark(lnv, 0);
-        for (i64 k = 0; k < e->ssz; k++) mark[sv[k]] = 1;
-        std::vector<unsigned> nperm(lnv);
-        i64 w = 0;
-        for (i64 s = 0; s < lnv; s++)
-            if (mark[perm_h[s]]) nperm[w++] = perm_h[s];
-        e->nexp = w;
-        for (i64 s = 0; s < lnv; s++)
-            if (!mark[perm_h[s]]) nperm[w++] = perm_h[s];
-        HIP_CHECK(hipMemcpyAsync(e->d_perm, nperm.data(), 4 * lnv,
-                                 hipMemcpyHostToDevice, st));
+        k_gather_flags<<<grid_for(lnv), 256, 0, st>>>(lnv, e->d_perm, d_mark,
+                                                      d_flags, 1);
+        (void)hipcub::DeviceSelect::Flagged(d_tsel, tsel, e->d_perm, d_flags,
+                                            d_out + e->nexp, d_num, lnv, st);
+        HIP_CHECK(hipMemcpyAsync(e->d_perm, d_out, 4 * lnv,
+                                 hipMemcpyDeviceToDevice, st));
         HIP_CHECK(hipStreamSynchronize(st));
+        HIP_CHECK(hipFree(d_mark));
+        HIP_CHECK(hipFree(d_flags));
+        HIP_CHECK(hipFree(d_out));
+        HIP_CHECK(hipFree(d_num));
+        HIP_CHECK(hipFree(d_tsel));
         e->overlap = 1;
     }
 
@@ -2238,12 +2281,8 @@ static void build_sell(mv_engine *e) {
         i64 total;
         if (!e->skewed) {
             const i64 per = std::max<i64>(e->max_degree, 1);
-            std::vector<i64> off(nthreads);
-            for (i64 t = 0; t < nthreads; t++) off[t] = t * per;
-            HIP_CHECK(hipMemcpyAsync(e->d_spill_off, off.data(),
-                                     8 * nthreads, hipMemcpyHostToDevice,
-                                     st));
-            HIP_CHECK(hipStreamSynchronize(st));
+            k_spill_uniform<<<grid_for(nthreads), 256, 0, st>>>(
+                nthreads, per, e->d_spill_off);
             total = nthreads * per;
         } else {
             i64 *d_need = nullptr;
@@ -2441,7 +2480,8 @@ double mv_engine_run(mv_engine *e, double lower, double thresh,
                                     8 * std::max<i64>((i64)nrem, 1)));
             size_t tmp1 = 0, tmp2 = 0;
             (void)hipcub::DeviceRadixSort::SortKeys(nullptr, tmp1, d_rem, d_sorted,
-                                              (int64_t)nrem, 0, 64, st);
+                                              (int64_t)nrem, 0, e->sort_bits,
+                                              st);
             i64 *d_ng = nullptr;
             HIP_CHECK(hipMalloc(&d_ng, 8));
             (void)hipcub::DeviceSelect::Unique(nullptr, tmp2, d_sorted, e->d_ghosts,
@@ -2450,7 +2490,8 @@ double mv_engine_run(mv_engine *e, double lower, double thresh,
             void *d_tmp = nullptr;
             HIP_CHECK(hipMalloc(&d_tmp, std::max<size_t>(tmpb, 1)));
             (void)hipcub::DeviceRadixSort::SortKeys(d_tmp, tmp1, d_rem, d_sorted,
-                                              (int64_t)nrem, 0, 64, st);
+                                              (int64_t)nrem, 0, e->sort_bits,
+                                              st);
             (void)hipcub::DeviceSelect::Unique(d_tmp, tmp2, d_sorted, e->d_ghosts,
                                          d_ng, (int64_t)nrem, st);
             HIP_CHECK(hipMemcpyAsync(&e->nghost, d_ng, 8,
@@ -2631,7 +2672,7 @@ double mv_engine_run(mv_engine *e, double lower, double thresh,
                 size_t t1 = 0, t2 = 0;
                 (void)hipcub::DeviceRadixSort::SortKeys(nullptr, t1, e->d_cand,
                                                   e->d_cand_sorted, cand_max, 0,
-                                                  64, st);
+                                                  e->sort_bits, st);
                 i64 *dummy = nullptr;
                 (void)hipcub::DeviceSelect::Unique(nullptr, t2, e->d_cand_sorted,
                                              e->d_rc_ids, dummy, cand_max, st);
@@ -2656,7 +2697,7 @@ double mv_engine_run(mv_engine *e, double lower, double thresh,
             size_t tb = e->cub_tmp_bytes;
             (void)hipcub::DeviceRadixSort::SortKeys(e->d_cub_tmp, tb, e->d_cand,
                                               e->d_cand_sorted, (int64_t)ncand,
-                                              0, 64, st);
+                                              0, e->sort_bits, st);
             i64 *d_nrc = (i64 *)e->d_count; // reuse as output slot
             tb = e->cub_tmp_bytes;
             (void)hipcub::DeviceSelect::Unique(e->d_cub_tmp, tb, e->d_cand_sorted,
