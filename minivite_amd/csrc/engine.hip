@@ -1623,7 +1623,10 @@ struct mv_engine {
     hipStream_t stream = nullptr;
     hipStream_t stream2 = nullptr; // halo #1a rides here, overlapped
     hipEvent_t ev_p1 = nullptr;    // sweep part 1 (exports) done [stream]
-    hipEvent_t ev_halo = nullptr;  // #1a recv complete [stream2]
+    hipEvent_t ev_p2 = nullptr;    // full sweep + writeback done [stream]
+    hipEvent_t ev_cinfo = nullptr; // K6 + delta application done [stream]
+    hipEvent_t ev_halo = nullptr;  // next-iteration pipeline done [stream2]
+    i64 *d_cnt_all = nullptr;      // persistent p*p counts buffer
     int overlap = 0;               // this run overlaps #1a (p>1, !skewed)
     i64 nexp = 0;                  // SELL positions [0,nexp) = exported
 
@@ -1786,6 +1789,8 @@ mv_engine *mv_engine_create(int device, int rank, int nranks,
     }
     HIP_CHECK(hipStreamCreate(&e->stream2));
     HIP_CHECK(hipEventCreate(&e->ev_p1));
+    HIP_CHECK(hipEventCreate(&e->ev_p2));
+    HIP_CHECK(hipEventCreate(&e->ev_cinfo));
     HIP_CHECK(hipEventCreate(&e->ev_halo));
     return e;
 }
@@ -1816,6 +1821,8 @@ mv_engine *mv_engine_create_lb(int device, int rank, int nranks,
     HIP_CHECK(hipStreamCreate(&e->stream));
     HIP_CHECK(hipStreamCreate(&e->stream2));
     HIP_CHECK(hipEventCreate(&e->ev_p1));
+    HIP_CHECK(hipEventCreate(&e->ev_p2));
+    HIP_CHECK(hipEventCreate(&e->ev_cinfo));
     HIP_CHECK(hipEventCreate(&e->ev_halo));
     return e;
 }
@@ -1825,6 +1832,8 @@ void mv_engine_destroy(mv_engine *e) {
     (void)hipSetDevice(e->device); // teardown: best effort
     for (auto ev : e->ev_pool) (void)hipEventDestroy(ev);
     if (e->ev_p1) (void)hipEventDestroy(e->ev_p1);
+    if (e->ev_p2) (void)hipEventDestroy(e->ev_p2);
+    if (e->ev_cinfo) (void)hipEventDestroy(e->ev_cinfo);
     if (e->ev_halo) (void)hipEventDestroy(e->ev_halo);
     if (e->comm2) ncclCommDestroy(e->comm2);
     if (e->comm) ncclCommDestroy(e->comm);
@@ -1863,7 +1872,7 @@ static void free_graph_state(mv_engine *e) {
                      (void **)&e->d_chg_lab, (void **)&e->d_chg_cnt,
                      (void **)&e->d_cntmat, (void **)&e->d_soff,
                      (void **)&e->d_ghost_labels, (void **)&e->d_rchg_idx,
-                     (void **)&e->d_rchg_lab,
+                     (void **)&e->d_rchg_lab, (void **)&e->d_cnt_all,
                      (void **)&e->d_trace_tmp}) {
         if (*p) {
             HIP_CHECK(hipFree(*p));
@@ -2065,7 +2074,8 @@ static void rccl_alltoallv(mv_engine *e, const void *send, const i64 *soff,
 
 // exchange per-peer counts: allgather of my nranks counts
 static void exchange_counts(mv_engine *e, const std::vector<i64> &mine,
-                            std::vector<i64> &matrix /*nranks*nranks*/) {
+                            std::vector<i64> &matrix /*nranks*nranks*/,
+                            ncclComm_t comm, hipStream_t stream) {
     matrix.resize((size_t)e->nranks * e->nranks);
     if (e->lb) {
         mv_lb_session *s = e->lb;
@@ -2077,16 +2087,18 @@ static void exchange_counts(mv_engine *e, const std::vector<i64> &mine,
         s->barrier();
         return;
     }
-    i64 *d_all = nullptr;
-    HIP_CHECK(hipMalloc(&d_all, 8 * e->nranks * e->nranks));
+    // persistent buffer: a hipMalloc/hipFree here would device-sync and
+    // re-serialize the overlapped pipeline
+    if (!e->d_cnt_all)
+        HIP_CHECK(hipMalloc(&e->d_cnt_all, 8 * e->nranks * e->nranks));
+    i64 *d_all = e->d_cnt_all;
     HIP_CHECK(hipMemcpyAsync(d_all + (i64)e->rank * e->nranks, mine.data(),
-                             8 * e->nranks, hipMemcpyHostToDevice, e->stream));
+                             8 * e->nranks, hipMemcpyHostToDevice, stream));
     NCCL_CHECK(ncclAllGather(d_all + (i64)e->rank * e->nranks, d_all,
-                             e->nranks, ncclInt64, e->comm, e->stream));
+                             e->nranks, ncclInt64, comm, stream));
     HIP_CHECK(hipMemcpyAsync(matrix.data(), d_all, 8 * e->nranks * e->nranks,
-                             hipMemcpyDeviceToHost, e->stream));
-    HIP_CHECK(hipStreamSynchronize(e->stream));
-    HIP_CHECK(hipFree(d_all));
+                             hipMemcpyDeviceToHost, stream));
+    HIP_CHECK(hipStreamSynchronize(stream));
 }
 
 // allreduce-sum of n (<= 2) host doubles in place (dspl.hpp:126, :441)
@@ -2344,7 +2356,7 @@ static void run_halo1a(mv_engine *e, const i64 *commArr, ncclComm_t comm,
             HIP_CHECK(hipMemcpyAsync(mine.data(), (i64 *)e->d_chg_cnt, 8 * p,
                                      hipMemcpyDeviceToHost, st2));
             HIP_CHECK(hipStreamSynchronize(st2));
-            exchange_counts(e, mine, m);
+            exchange_counts(e, mine, m, comm, st2);
             std::copy(m.begin(), m.end(), e->h_cntmat);
         } else {
             HIP_CHECK(hipMemcpyAsync(e->d_cntmat + (i64)e->rank * p,
@@ -2439,6 +2451,116 @@ static void run_halo1a(mv_engine *e, const i64 *commArr, ncclComm_t comm,
     }
 }
 
+// Candidate discovery + halo #1b/#1c/#1d + u32 view build for ONE
+// iteration, on (comm, st2): collect the remote communities referenced by
+// `curr` / the ghost labels, fetch their (size, degree) records from the
+// owners (dspl.hpp:670-929), zero rcu, and build the sweep views. In
+// overlap mode this runs on stream2/comm2 during the PREVIOUS iteration's
+// epilogue; `ev_cinfo` (when given) gates the reply section on that
+// iteration's cinfo updates (K6 + per-sender delta application) — the
+// replies must see exactly the reference's post-update localCinfo.
+static void halo_prep(mv_engine *e, const i64 *curr,
+                      std::vector<i64> &rc_bounds, std::vector<i64> &req_off,
+                      i64 &nrc_out, ncclComm_t comm, hipStream_t st2,
+                      hipEvent_t ev_cinfo) {
+    const int p = e->nranks, me = e->rank;
+    const i64 lnv = e->lnv;
+    // ---- needed remote communities (dspl.hpp:670-700) ----
+    const i64 cand_max = e->nghost + lnv;
+    if (cand_max > e->rc_cap) {
+        for (void *q : {(void *)e->d_cand, (void *)e->d_cand_sorted,
+                        (void *)e->d_rc_ids, (void *)e->d_rc_info,
+                        (void *)e->d_rcu})
+            if (q) HIP_CHECK(hipFree(q));
+        e->rc_cap = cand_max;
+        HIP_CHECK(hipMalloc(&e->d_cand, 8 * cand_max));
+        HIP_CHECK(hipMalloc(&e->d_cand_sorted, 8 * cand_max));
+        HIP_CHECK(hipMalloc(&e->d_rc_ids, 8 * cand_max));
+        HIP_CHECK(hipMalloc(&e->d_rc_info, sizeof(Info16) * cand_max));
+        HIP_CHECK(hipMalloc(&e->d_rcu, sizeof(Info16) * cand_max));
+        size_t t1 = 0, t2 = 0;
+        (void)hipcub::DeviceRadixSort::SortKeys(nullptr, t1, e->d_cand,
+                                                e->d_cand_sorted, cand_max,
+                                                0, e->sort_bits, st2);
+        i64 *dummy = nullptr;
+        (void)hipcub::DeviceSelect::Unique(nullptr, t2, e->d_cand_sorted,
+                                           e->d_rc_ids, dummy, cand_max,
+                                           st2);
+        size_t need = std::max(t1, t2);
+        if (need > e->cub_tmp_bytes) {
+            if (e->d_cub_tmp) HIP_CHECK(hipFree(e->d_cub_tmp));
+            HIP_CHECK(hipMalloc(&e->d_cub_tmp, need));
+            e->cub_tmp_bytes = need;
+        }
+    }
+    HIP_CHECK(hipMemsetAsync(e->d_count, 0, 8, st2));
+    k_filter_remote<<<grid_for(std::max<i64>(e->nghost, 1)), 256, 0, st2>>>(
+        e->nghost, e->d_ghost_labels, /*shift*/ 0, e->base, e->bound,
+        e->d_cand, e->d_count);
+    k_filter_remote<<<grid_for(lnv), 256, 0, st2>>>(
+        lnv, curr, /*shift*/ 0, e->base, e->bound, e->d_cand, e->d_count);
+    unsigned long long ncand = 0;
+    HIP_CHECK(hipMemcpyAsync(&ncand, e->d_count, 8, hipMemcpyDeviceToHost,
+                             st2));
+    HIP_CHECK(hipStreamSynchronize(st2));
+    size_t tb = e->cub_tmp_bytes;
+    (void)hipcub::DeviceRadixSort::SortKeys(e->d_cub_tmp, tb, e->d_cand,
+                                            e->d_cand_sorted, (int64_t)ncand,
+                                            0, e->sort_bits, st2);
+    i64 *d_nrc = (i64 *)e->d_count; // reuse as output slot
+    tb = e->cub_tmp_bytes;
+    (void)hipcub::DeviceSelect::Unique(e->d_cub_tmp, tb, e->d_cand_sorted,
+                                       e->d_rc_ids, d_nrc, (int64_t)ncand,
+                                       st2);
+    i64 nrc = 0;
+    HIP_CHECK(hipMemcpyAsync(&nrc, d_nrc, 8, hipMemcpyDeviceToHost, st2));
+    HIP_CHECK(hipStreamSynchronize(st2));
+
+    // ---- halo #1b/#1c/#1d: request (size,degree) of those communities
+    // from their owners (dspl.hpp:719-929) ----
+    k_owner_bounds<<<1, p + 1, 0, st2>>>(e->d_rc_ids, nrc, e->d_parts, p,
+                                         e->d_bounds);
+    rc_bounds.assign(p + 1, 0);
+    HIP_CHECK(hipMemcpyAsync(rc_bounds.data(), e->d_bounds, 8 * (p + 1),
+                             hipMemcpyDeviceToHost, st2));
+    HIP_CHECK(hipStreamSynchronize(st2));
+    std::vector<i64> reqs(p), matrix;
+    for (int r = 0; r < p; r++) reqs[r] = rc_bounds[r + 1] - rc_bounds[r];
+    exchange_counts(e, reqs, matrix, comm, st2);
+    req_off.assign(p + 1, 0);
+    for (int r = 0; r < p; r++)
+        req_off[r + 1] =
+            req_off[r] + ((r == me) ? 0 : matrix[(size_t)r * p + me]);
+    const i64 nreq = req_off[p];
+    if (nreq > e->req_cap) { // rare growth; device-wide hipFree sync is ok
+        if (e->d_req_ids) HIP_CHECK(hipFree(e->d_req_ids));
+        if (e->d_req_info) HIP_CHECK(hipFree(e->d_req_info));
+        e->req_cap = std::max<i64>(nreq, 64);
+        HIP_CHECK(hipMalloc(&e->d_req_ids, 8 * e->req_cap));
+        HIP_CHECK(hipMalloc(&e->d_req_info, sizeof(Info16) * e->req_cap));
+    }
+    // the previous iteration's delta routing reads d_req_ids/d_req_info
+    // and writes cinfo; #1c's recv and the replies must order after it
+    if (ev_cinfo) HIP_CHECK(hipStreamWaitEvent(st2, ev_cinfo, 0));
+    rccl_alltoallv(e, e->d_rc_ids, rc_bounds.data(), e->d_req_ids,
+                   req_off.data(), 8, ncclInt64, 8, comm, st2);
+    k9_reply_info<<<grid_for(std::max<i64>(nreq, 1)), 256, 0, st2>>>(
+        nreq, e->d_req_ids, e->base, e->d_sigma_inv, e->d_cinfo,
+        e->d_req_info);
+    rccl_alltoallv(e, e->d_req_info, req_off.data(), e->d_rc_info,
+                   rc_bounds.data(), sizeof(Info16), ncclChar, 1, comm, st2);
+    HIP_CHECK(hipMemsetAsync(e->d_rcu, 0,
+                             sizeof(Info16) * std::max<i64>(nrc, 1), st2));
+    // u32 views for the sweep (slot / lnv + rc index encoding)
+    k_build_view<<<grid_for(lnv), 256, 0, st2>>>(
+        lnv, curr, e->base, e->bound, lnv, e->d_sigma_inv, e->d_rc_ids, nrc,
+        e->d_vcurr);
+    k_build_view<<<grid_for(std::max<i64>(e->nghost, 1)), 256, 0, st2>>>(
+        e->nghost, e->d_ghost_labels, e->base, e->bound, lnv,
+        e->d_sigma_inv, e->d_rc_ids, nrc, e->d_vghost);
+    nrc_out = nrc;
+}
+
 #define PHASE(tag)                                                            \
     do {                                                                      \
         if (getenv("MV_PHASE_DEBUG")) {                                       \
@@ -2514,7 +2636,7 @@ double mv_engine_run(mv_engine *e, double lower, double thresh,
             std::vector<i64> want(p), matrix;
             for (int r = 0; r < p; r++)
                 want[r] = e->recv_off[r + 1] - e->recv_off[r];
-            exchange_counts(e, want, matrix);
+            exchange_counts(e, want, matrix, e->comm, e->stream);
             e->send_off.assign(p + 1, 0);
             for (int r = 0; r < p; r++)
                 e->send_off[r + 1] =
@@ -2583,7 +2705,7 @@ double mv_engine_run(mv_engine *e, double lower, double thresh,
             // GLOBAL decision: all ranks take it or none (a rank can be
             // locally skewed / export-free while others are not)
             std::vector<i64> flag(p, e->overlap ? 1 : 0), fm;
-            exchange_counts(e, flag, fm);
+            exchange_counts(e, flag, fm, e->comm, e->stream);
             for (int r = 0; r < p; r++)
                 if (fm[(size_t)r * p] == 0) e->overlap = 0;
         }
@@ -2629,16 +2751,25 @@ double mv_engine_run(mv_engine *e, double lower, double thresh,
     i64 *d_curr = e->d_curr, *d_past = e->d_past, *d_target = e->d_target;
     double prevMod = lower, currMod = -1.0;
     int numIters = 0;
+    // halo state for the CURRENT iteration, and the one the overlapped
+    // pipeline prepares for the NEXT (swapped at each loop top)
     std::vector<i64> rc_bounds(p + 1, 0), req_off(p + 1, 0);
+    std::vector<i64> rc_bounds_nx(p + 1, 0), req_off_nx(p + 1, 0);
+    i64 nrc = 0, nrc_nx = 0;
 
-    // Overlap mode: run the NEXT iteration's halo #1a on stream2/comm2 as
-    // soon as the export positions' targets are written (sweep part 1),
-    // hiding the exchange behind the interior sweep and the epilogue.
-    // Joined via ev_halo at the next iteration's top.
+    // Overlap mode: the ENTIRE next-iteration halo pipeline — #1a ghost
+    // communities off the exports' targets, candidate discovery, counts,
+    // #1c/#1d info fetch, rcu zero and the view builds — runs on
+    // stream2/comm2 while stream A sweeps the interior and runs the
+    // epilogue; only #2 + the modularity allreduce stay on the critical
+    // path. Joined via ev_halo at the next iteration's top.
+    ncclComm_t c2 = e->comm2 ? e->comm2 : e->comm;
     if (p > 1 && e->overlap) {
         HIP_CHECK(hipEventRecord(e->ev_p1, st));
         HIP_CHECK(hipStreamWaitEvent(e->stream2, e->ev_p1, 0));
-        run_halo1a(e, e->d_curr, e->comm2 ? e->comm2 : e->comm, e->stream2);
+        run_halo1a(e, e->d_curr, c2, e->stream2);
+        halo_prep(e, e->d_curr, rc_bounds, req_off, nrc, c2, e->stream2,
+                  nullptr); // initial cinfo (K1) is complete: A was synced
         HIP_CHECK(hipEventRecord(e->ev_halo, e->stream2));
     }
 
@@ -2646,111 +2777,23 @@ double mv_engine_run(mv_engine *e, double lower, double thresh,
     for (;;) {
         numIters++;
 
-        i64 nrc = 0;
         if (p > 1) {
             const auto t_h0 = std::chrono::steady_clock::now();
-            // ---- halo #1a: ghost communities (dspl.hpp:583-647) ----
             if (e->overlap) {
+                // pipeline prepared everything during the previous
+                // iteration's epilogue (or pre-loop)
                 HIP_CHECK(hipStreamWaitEvent(st, e->ev_halo, 0));
+                if (numIters > 1) {
+                    std::swap(rc_bounds, rc_bounds_nx);
+                    std::swap(req_off, req_off_nx);
+                    nrc = nrc_nx;
+                }
             } else {
                 run_halo1a(e, d_curr, e->comm, st);
+                halo_prep(e, d_curr, rc_bounds, req_off, nrc, e->comm, st,
+                          nullptr);
+                HIP_CHECK(hipStreamSynchronize(st));
             }
-
-            // ---- needed remote communities (dspl.hpp:670-700) ----
-            const i64 cand_max = e->nghost + lnv;
-            if (cand_max > e->rc_cap) {
-                for (void *q : {(void *)e->d_cand, (void *)e->d_cand_sorted,
-                                (void *)e->d_rc_ids, (void *)e->d_rc_info,
-                                (void *)e->d_rcu})
-                    if (q) HIP_CHECK(hipFree(q));
-                e->rc_cap = cand_max;
-                HIP_CHECK(hipMalloc(&e->d_cand, 8 * cand_max));
-                HIP_CHECK(hipMalloc(&e->d_cand_sorted, 8 * cand_max));
-                HIP_CHECK(hipMalloc(&e->d_rc_ids, 8 * cand_max));
-                HIP_CHECK(hipMalloc(&e->d_rc_info, sizeof(Info16) * cand_max));
-                HIP_CHECK(hipMalloc(&e->d_rcu, sizeof(Info16) * cand_max));
-                size_t t1 = 0, t2 = 0;
-                (void)hipcub::DeviceRadixSort::SortKeys(nullptr, t1, e->d_cand,
-                                                  e->d_cand_sorted, cand_max, 0,
-                                                  e->sort_bits, st);
-                i64 *dummy = nullptr;
-                (void)hipcub::DeviceSelect::Unique(nullptr, t2, e->d_cand_sorted,
-                                             e->d_rc_ids, dummy, cand_max, st);
-                size_t need = std::max(t1, t2);
-                if (need > e->cub_tmp_bytes) {
-                    if (e->d_cub_tmp) HIP_CHECK(hipFree(e->d_cub_tmp));
-                    HIP_CHECK(hipMalloc(&e->d_cub_tmp, need));
-                    e->cub_tmp_bytes = need;
-                }
-            }
-            HIP_CHECK(hipMemsetAsync(e->d_count, 0, 8, st));
-            k_filter_remote<<<grid_for(std::max<i64>(e->nghost, 1)), 256, 0,
-                              st>>>(e->nghost, e->d_ghost_labels, /*shift*/ 0,
-                                    e->base, e->bound, e->d_cand, e->d_count);
-            k_filter_remote<<<grid_for(lnv), 256, 0, st>>>(
-                lnv, d_curr, /*shift*/ 0, e->base, e->bound, e->d_cand,
-                e->d_count);
-            unsigned long long ncand = 0;
-            HIP_CHECK(hipMemcpyAsync(&ncand, e->d_count, 8,
-                                     hipMemcpyDeviceToHost, st));
-            HIP_CHECK(hipStreamSynchronize(st));
-            size_t tb = e->cub_tmp_bytes;
-            (void)hipcub::DeviceRadixSort::SortKeys(e->d_cub_tmp, tb, e->d_cand,
-                                              e->d_cand_sorted, (int64_t)ncand,
-                                              0, e->sort_bits, st);
-            i64 *d_nrc = (i64 *)e->d_count; // reuse as output slot
-            tb = e->cub_tmp_bytes;
-            (void)hipcub::DeviceSelect::Unique(e->d_cub_tmp, tb, e->d_cand_sorted,
-                                         e->d_rc_ids, d_nrc, (int64_t)ncand, st);
-            HIP_CHECK(hipMemcpyAsync(&nrc, d_nrc, 8, hipMemcpyDeviceToHost, st));
-            // (labels -> handles conversion happened inside run_halo1a)
-            HIP_CHECK(hipStreamSynchronize(st));
-
-            // ---- halo #1b/#1c/#1d: request (size,degree) of those
-            // communities from their owners (dspl.hpp:719-929) ----
-            k_owner_bounds<<<1, p + 1, 0, st>>>(e->d_rc_ids, nrc, e->d_parts,
-                                                p, e->d_bounds);
-            HIP_CHECK(hipMemcpyAsync(rc_bounds.data(), e->d_bounds, 8 * (p + 1),
-                                     hipMemcpyDeviceToHost, st));
-            HIP_CHECK(hipStreamSynchronize(st));
-            std::vector<i64> reqs(p), matrix;
-            for (int r = 0; r < p; r++)
-                reqs[r] = rc_bounds[r + 1] - rc_bounds[r];
-            exchange_counts(e, reqs, matrix);
-            req_off[0] = 0;
-            for (int r = 0; r < p; r++)
-                req_off[r + 1] =
-                    req_off[r] + ((r == me) ? 0 : matrix[(size_t)r * p + me]);
-            const i64 nreq = req_off[p];
-            if (nreq > e->req_cap) {
-                if (e->d_req_ids) HIP_CHECK(hipFree(e->d_req_ids));
-                if (e->d_req_info) HIP_CHECK(hipFree(e->d_req_info));
-                e->req_cap = std::max<i64>(nreq, 64);
-                HIP_CHECK(hipMalloc(&e->d_req_ids, 8 * e->req_cap));
-                HIP_CHECK(
-                    hipMalloc(&e->d_req_info, sizeof(Info16) * e->req_cap));
-            }
-            rccl_alltoallv(e, e->d_rc_ids, rc_bounds.data(), e->d_req_ids,
-                           req_off.data(), 8, ncclInt64, 8, e->comm,
-                           e->stream);
-            k9_reply_info<<<grid_for(std::max<i64>(nreq, 1)), 256, 0, st>>>(
-                nreq, e->d_req_ids, e->base, e->d_sigma_inv, e->d_cinfo,
-                e->d_req_info);
-            rccl_alltoallv(e, e->d_req_info, req_off.data(), e->d_rc_info,
-                           rc_bounds.data(), sizeof(Info16), ncclChar, 1,
-                           e->comm, e->stream);
-            HIP_CHECK(hipMemsetAsync(e->d_rcu, 0,
-                                     sizeof(Info16) * std::max<i64>(nrc, 1),
-                                     st));
-            // u32 views for the sweep (slot / lnv + rc index encoding)
-            k_build_view<<<grid_for(lnv), 256, 0, st>>>(
-                lnv, d_curr, e->base, e->bound, lnv, e->d_sigma_inv,
-                e->d_rc_ids, nrc, e->d_vcurr);
-            k_build_view<<<grid_for(std::max<i64>(e->nghost, 1)), 256, 0,
-                           st>>>(e->nghost, e->d_ghost_labels, e->base,
-                                 e->bound, lnv, e->d_sigma_inv, e->d_rc_ids,
-                                 nrc, e->d_vghost);
-            HIP_CHECK(hipStreamSynchronize(st));
             e->stats.halo_ms +=
                 std::chrono::duration<double, std::milli>(
                     std::chrono::steady_clock::now() - t_h0)
@@ -2902,18 +2945,16 @@ double mv_engine_run(mv_engine *e, double lower, double thresh,
                 e->d_rc_ids, lnv, d_target);
         };
         if (p > 1 && e->overlap) {
-            // exports first (sweep + label writeback, so halo #1a can
-            // gather them); the interior launch goes in BEFORE the halo's
-            // host-side work so it runs under the exchange
+            // exports first (their targets feed the next iteration's #1a
+            // once ev_p1 fires); the full next-iteration pipeline is
+            // issued after the epilogue's delta application (ev_cinfo) so
+            // it runs under the epilogue + exit check — see below
             sweep_range(0, e->nexp);
             writeback(0, e->nexp);
             HIP_CHECK(hipEventRecord(e->ev_p1, st));
             sweep_range(e->nexp, lnv);
             writeback(e->nexp, lnv);
-            HIP_CHECK(hipStreamWaitEvent(e->stream2, e->ev_p1, 0));
-            run_halo1a(e, d_target, e->comm2 ? e->comm2 : e->comm,
-                       e->stream2);
-            HIP_CHECK(hipEventRecord(e->ev_halo, e->stream2));
+            HIP_CHECK(hipEventRecord(e->ev_p2, st));
         } else {
             sweep_range(e->nhi, lnv);
             writeback(0, lnv);
@@ -2951,7 +2992,13 @@ double mv_engine_run(mv_engine *e, double lower, double thresh,
                         c, e->d_req_ids + req_off[r], e->base, e->d_sigma_inv,
                         e->d_req_info + req_off[r], e->d_cinfo);
             }
-            HIP_CHECK(hipStreamSynchronize(st));
+            if (e->overlap) {
+                // the next-iteration pipeline's replies must see THIS
+                // post-update cinfo and may reuse the req buffers after it
+                HIP_CHECK(hipEventRecord(e->ev_cinfo, st));
+            } else {
+                HIP_CHECK(hipStreamSynchronize(st));
+            }
             e->stats.halo_ms +=
                 std::chrono::duration<double, std::milli>(
                     std::chrono::steady_clock::now() - t_h0)
@@ -2972,6 +3019,18 @@ double mv_engine_run(mv_engine *e, double lower, double thresh,
         PHASE("k67-done");
         HIP_CHECK(hipMemcpyAsync(partials, e->d_partials, 16 * nblocks,
                                  hipMemcpyDeviceToHost, st));
+        if (p > 1 && e->overlap) {
+            // issue the WHOLE next-iteration halo pipeline on stream2 now
+            // (A's epilogue is queued and drains concurrently): #1a off
+            // the exports (ev_p1), discovery off the full targets (ev_p2),
+            // replies off the updated cinfo (ev_cinfo)
+            HIP_CHECK(hipStreamWaitEvent(e->stream2, e->ev_p1, 0));
+            run_halo1a(e, d_target, c2, e->stream2);
+            HIP_CHECK(hipStreamWaitEvent(e->stream2, e->ev_p2, 0));
+            halo_prep(e, d_target, rc_bounds_nx, req_off_nx, nrc_nx, c2,
+                      e->stream2, e->ev_cinfo);
+            HIP_CHECK(hipEventRecord(e->ev_halo, e->stream2));
+        }
         HIP_CHECK(hipStreamSynchronize(st));
         double le = 0.0, la = 0.0;
         for (int b = 0; b < nblocks; b++) {
