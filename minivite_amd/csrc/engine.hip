@@ -29,9 +29,9 @@
 //    lookup (dspl.hpp:253-260). Unit-weight graphs skip the weight stream
 //    entirely (every w == 1.0, detected at load).
 //  * Community info (localCinfo/localCupdate, dspl.hpp:61-66) is AoS
-//    {int64 size; double degree} in INTERNAL order, addressed through
-//    packed community handles (see h_label/h_slot below), so one lookup
-//    costs one spatially-local cache line.
+//    {int64 size; double degree} in INTERNAL order. Persistent community
+//    arrays carry LABELS; sweeps run over u32 slot (p==1) / view (p>1)
+//    encodings that index cinfo/rc_info directly (see k_build_view).
 //
 // FP discipline: built with -ffp-contract=off so the dQ gain expression
 // (dspl.hpp:212) and all accumulations carry the same bits as the
@@ -93,25 +93,12 @@ struct Info16 {    // wire record for cinfo replies / delta routing: the
     double degree; // id, which both ends know by position
 };
 
-// Community HANDLES: community values inside the engine are
-// (label << 32) | slot, where label is the reference's community id
-// (a global vertex id) and slot is the owner-local INTERNAL index of the
-// label's home vertex (valid only for locally-owned labels; remote
-// handles carry slot 0 and are resolved through the sorted rc_ids
-// label array). The label occupies the high bits, so handle comparison
-// and equality reproduce the reference's label comparison exactly
-// (dspl.hpp:214-215 tie-break included); localCinfo/localCupdate live in
-// INTERNAL order, making candidate-info gathers as spatially local as the
-// community structure itself. Labels on the wire stay plain (handles are
-// never exchanged). Requires nv < 2^31 (checked at load).
-__device__ __forceinline__ i64 h_label(i64 h) { return h >> 32; }
-// Speculation-safe index clamps: the branches below ARE semantically
-// exclusive, but the compiler may if-convert them into two unconditional
-// loads + a value select, so the un-taken side's address must still be
-// dereferenceable (ghost/rc buffers are never null — load_graph keeps
-// 1-element dummies — and indices are clamped non-negative).
+// Speculation-safe index clamp: semantically-exclusive branches may be
+// if-converted into two unconditional loads + a value select, so the
+// un-taken side's address must still be dereferenceable (ghost/rc
+// buffers are never null — load_graph keeps 1-element dummies — and
+// indices are clamped non-negative).
 __device__ __forceinline__ i64 clamp0(i64 x) { return x > 0 ? x : 0; }
-__device__ __forceinline__ i64 h_slot(i64 h) { return h & 0xFFFFFFFFll; }
 
 __device__ __forceinline__ i64 dev_lower_bound(const i64 *a, i64 n, i64 key) {
     i64 lo = 0, hi = n;
@@ -1907,7 +1894,8 @@ int mv_engine_load_graph(mv_engine *e, const mv_graph *g) {
     HIP_CHECK(hipSetDevice(e->device));
     if (e->d_xadj) free_graph_state(e); // re-load on a live engine
     e->nv = mv_graph_nv(g);
-    if (e->nv >= (1ll << 31)) { // community-handle packing needs 31-bit ids
+    if (e->nv >= (1ll << 31)) { // u32 slot/view encodings + int32 SELL
+                                // tails need 31-bit ids
         std::fprintf(stderr, "mv_engine_load_graph: nv >= 2^31 unsupported\n");
         return -1;
     }
