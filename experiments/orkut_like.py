@@ -22,7 +22,12 @@ REPO = os.path.dirname(os.path.dirname(os.path.abspath(__file__)))
 sys.path.insert(0, REPO)
 
 
-def build_graph(nv, avg_deg, max_deg, seed=1):
+def build_graph(nv, avg_deg, max_deg, seed=1, structure=True,
+                p_intra=0.6, ncomm=100000):
+    """Power-law-degree graph; with `structure`, ~p_intra of each vertex's
+    edges go to members of its planted community (Zipf community sizes) —
+    real clustering like a social graph, so Louvain actually iterates and
+    candidate counts per vertex stay small after the first rounds."""
     rng = np.random.default_rng(seed)
     # Zipf-ish out-degrees scaled to hit avg_deg/2 before symmetrization
     raw = (1.0 / rng.power(1.6, nv)).astype(np.int64)
@@ -30,7 +35,22 @@ def build_graph(nv, avg_deg, max_deg, seed=1):
     scale = (avg_deg / 2.0) / raw.mean()
     deg = np.maximum((raw * scale).astype(np.int64), 1)
     src = np.repeat(np.arange(nv, dtype=np.int64), deg)
-    dst = rng.integers(0, nv, src.size, dtype=np.int64)
+    if structure:
+        # planted communities with Zipf sizes
+        cw = 1.0 / np.arange(1, ncomm + 1) ** 0.9
+        comm = rng.choice(ncomm, nv, p=cw / cw.sum())
+        order_v = np.argsort(comm, kind="stable")
+        cstart = np.searchsorted(comm[order_v], np.arange(ncomm))
+        cend = np.searchsorted(comm[order_v], np.arange(ncomm), side="right")
+        csz = cend - cstart
+        intra = rng.random(src.size) < p_intra
+        cs = comm[src]
+        # random member of src's community (may self-hit; filtered below)
+        pick = cstart[cs] + (rng.random(src.size) * csz[cs]).astype(np.int64)
+        dst = np.where(intra, order_v[np.minimum(pick, len(order_v) - 1)],
+                       rng.integers(0, nv, src.size, dtype=np.int64))
+    else:
+        dst = rng.integers(0, nv, src.size, dtype=np.int64)
     keep = src != dst
     src, dst = src[keep], dst[keep]
     u = np.concatenate([src, dst])
@@ -50,12 +70,16 @@ def main():
     ap.add_argument("--max-deg", type=int, default=33000)
     ap.add_argument("--steps", type=int, default=3)
     ap.add_argument("--warmup", type=int, default=1)
+    ap.add_argument("--random", action="store_true",
+                    help="no planted communities (worst-case stress: every "
+                         "neighbor a distinct candidate)")
     args = ap.parse_args()
 
     from minivite_amd import Graph, Engine
 
     t0 = time.perf_counter()
-    xadj, tails = build_graph(args.nv, args.avg_deg, args.max_deg)
+    xadj, tails = build_graph(args.nv, args.avg_deg, args.max_deg,
+                              structure=not args.random)
     gen_s = time.perf_counter() - t0
     lne = len(tails)
 
@@ -84,7 +108,8 @@ def main():
     g.free()
     print(json.dumps({
         "workload": f"orkut_like_nv{args.nv}_deg{args.avg_deg}"
-                    f"_cap{args.max_deg}",
+                    f"_cap{args.max_deg}"
+                    f"_{'random' if args.random else 'planted'}",
         "directed_edges": lne,
         "max_degree": int((xadj[1:] - xadj[:-1]).max()),
         "iterations": iters,
