@@ -2147,7 +2147,13 @@ static void build_sell(mv_engine *e) {
         HIP_CHECK(hipMemcpyAsync(sdeg.data(), d_degs, 4 * lnv,
                                  hipMemcpyDeviceToHost, st));
         HIP_CHECK(hipStreamSynchronize(st));
-        constexpr unsigned HI_THRESH = 256;
+        // MV_HI_THRESH (perf only): degree bound above which a vertex
+        // takes the hub path (wave hash for unit, per-lane hash for -w)
+        // instead of the LDS-slot + linear-spill lane path
+        static const unsigned HI_THRESH = [] {
+            const char *s = getenv("MV_HI_THRESH");
+            return s ? (unsigned)atoi(s) : 256u;
+        }();
         i64 nhi = 0;
         while (nhi < lnv && sdeg[nhi] > HI_THRESH) nhi++;
         e->nhi = e->skewed ? nhi : 0;
