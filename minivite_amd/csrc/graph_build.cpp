@@ -103,36 +103,43 @@ static void build_bfs_hint(mv_graph *g) {
     const int64_t lnv = (int64_t)g->xadj.size() - 1;
     if (lnv <= 1) return;
     const int64_t base = g->parts[g->rank], bound = g->parts[g->rank + 1];
-    std::vector<int32_t> seeds(lnv);
-    for (int64_t i = 0; i < lnv; i++) seeds[i] = (int32_t)i;
-    std::stable_sort(seeds.begin(), seeds.end(),
-                     [&](int32_t a, int32_t b) {
-                         return g->xadj[a + 1] - g->xadj[a] >
-                                g->xadj[b + 1] - g->xadj[b];
-                     });
-    std::vector<char> vis(lnv, 0);
-    std::vector<int32_t> q;
-    q.reserve(lnv);
-    size_t qh = 0;
-    for (int64_t s = 0; s < lnv; s++) {
-        const int32_t seed = seeds[s];
-        if (vis[seed]) continue;
-        vis[seed] = 1;
-        q.push_back(seed);
-        while (qh < q.size()) {
-            const int32_t v = q[qh++];
+    // 2 rounds of async label propagation over the LOCAL subgraph
+    // (deterministic sequential order; timestamped counters keep it
+    // O(edges)); vertices sharing a propagated label end up adjacent, so
+    // a vertex's neighbors — mostly co-members once Louvain coarsens —
+    // sit in the same cache blocks.
+    std::vector<int32_t> label(lnv), stamp(lnv, -1), cnt(lnv, 0);
+    for (int64_t i = 0; i < lnv; i++) label[i] = (int32_t)i;
+    for (int round = 0; round < 2; round++) {
+        for (int64_t v = 0; v < lnv; v++) {
+            int32_t best = label[v];
+            int32_t bestc = 0;
             for (int64_t e = g->xadj[v]; e < g->xadj[v + 1]; e++) {
                 const int64_t t = g->tails[e];
                 if (t < base || t >= bound) continue;
-                const int64_t tl = t - base;
-                if (!vis[tl]) {
-                    vis[tl] = 1;
-                    q.push_back((int32_t)tl);
+                const int32_t l = label[t - base];
+                if (stamp[l] != (int32_t)v) {
+                    stamp[l] = (int32_t)v;
+                    cnt[l] = 0;
+                }
+                const int32_t c = ++cnt[l];
+                if (c > bestc || (c == bestc && l < best)) {
+                    bestc = c;
+                    best = l;
                 }
             }
+            label[v] = best;
         }
+        std::fill(stamp.begin(), stamp.end(), -1);
     }
-    g->locality_perm = std::move(q);
+    // order = (label block, id); stable so ties keep id order
+    std::vector<int32_t> order(lnv);
+    for (int64_t i = 0; i < lnv; i++) order[i] = (int32_t)i;
+    std::stable_sort(order.begin(), order.end(),
+                     [&](int32_t a, int32_t b) {
+                         return label[a] < label[b];
+                     });
+    g->locality_perm = std::move(order);
 }
 
 extern "C" {
