@@ -336,6 +336,24 @@ __global__ void k_spill_need(i64 nthreads, i64 s_begin, i64 lnv,
     }
 }
 
+// translate the full CSR tail array into int32 internal indices (local ->
+// internal slot, ghost -> lnv + ghost index) once at build: the hub wave
+// kernels walk CSR rows and would otherwise pay sigma_inv[tails[e]] — two
+// dependent random gathers — per edge per iteration
+__global__ void k_translate_tails(i64 lne, const i64 *__restrict__ tails,
+                                  i64 base, i64 bound,
+                                  const unsigned *__restrict__ sigma_inv,
+                                  const i64 *__restrict__ ghosts, i64 nghost,
+                                  i64 lnv, int *__restrict__ out) {
+    for (i64 e = blockIdx.x * (i64)blockDim.x + threadIdx.x; e < lne;
+         e += (i64)gridDim.x * blockDim.x) {
+        const i64 t = tails[e];
+        out[e] = (t >= base && t < bound)
+                     ? (int)sigma_inv[t - base]
+                     : (int)(lnv + dev_lower_bound(ghosts, nghost, t));
+    }
+}
+
 // translate a global-id list into internal indices (svdata, once per run)
 __global__ void k_to_internal(i64 n, const i64 *__restrict__ gids, i64 base,
                               const unsigned *__restrict__ sigma_inv,
@@ -736,10 +754,9 @@ __global__ __launch_bounds__(256) void k4_sweep_mr(
 // accumulation order; -w skewed graphs take the serial lane path instead
 // (edge-order bit parity).
 __global__ __launch_bounds__(256) void k4_sweep_hi_mr(
-    i64 nhi, i64 lnv, i64 base, i64 bound, const unsigned *__restrict__ perm,
+    i64 nhi, i64 lnv, i64 base, const unsigned *__restrict__ perm,
     const unsigned *__restrict__ deg_int, const unsigned *__restrict__ sigma,
-    const unsigned *__restrict__ sigma_inv, const i64 *__restrict__ xadj,
-    const i64 *__restrict__ tails, const i64 *__restrict__ ghosts, i64 nghost,
+    const i64 *__restrict__ xadj, const int *__restrict__ tidx32,
     const unsigned *__restrict__ vcurr, const unsigned *__restrict__ vghost,
     const double *__restrict__ vDegree, const Cinfo *__restrict__ cinfo,
     Cinfo *__restrict__ cupd, const i64 *__restrict__ rc_ids,
@@ -762,7 +779,6 @@ __global__ __launch_bounds__(256) void k4_sweep_hi_mr(
         const unsigned cc = vcurr[i];
         const i64 hoff = hash_off[s];
         const i64 cap = hash_off[s + 1] - hoff; // power of two
-        const i64 vglobal = v + base;
         double ccDeg;
         i64 ccSize;
         if (cc < lnv) {
@@ -776,12 +792,9 @@ __global__ __launch_bounds__(256) void k4_sweep_hi_mr(
         }
         double c0 = 0.0, selfLoop = 0.0;
         for (int k = lane; k < deg; k += 64) {
-            const i64 tail = tails[e0 + k];
-            const double w = 1.0; // hub path is unit-only
-            if (tail == vglobal) selfLoop += w;
-            const i64 ti = (tail >= base && tail < bound)
-                               ? (i64)sigma_inv[tail - base]
-                               : lnv + dev_lower_bound(ghosts, nghost, tail);
+            const i64 ti = tidx32[e0 + k]; // pre-translated internal index
+            const double w = 1.0;          // hub path is unit-only
+            if (ti == i) selfLoop += w;
             const unsigned tcomm = (ti < lnv) ? vcurr[ti]
                                               : vghost[ti - lnv];
             if (tcomm == cc) { c0 += w; continue; }
@@ -1452,8 +1465,8 @@ __global__ __launch_bounds__(256) void k4_sweep_iter1_p1(
 __global__ __launch_bounds__(256) void k4_sweep_hi_p1(
     i64 nhi, i64 lnv, const unsigned *__restrict__ perm,
     const unsigned *__restrict__ deg_int, const unsigned *__restrict__ sigma,
-    const unsigned *__restrict__ sigma_inv, const i64 *__restrict__ xadj,
-    const i64 *__restrict__ tails, const unsigned *__restrict__ currComm,
+    const i64 *__restrict__ xadj,
+    const int *__restrict__ tidx32, const unsigned *__restrict__ currComm,
     const double *__restrict__ vDegree, const Cinfo *__restrict__ cinfo,
     Cinfo *__restrict__ cupd, double constant,
     unsigned *__restrict__ targetComm, double *__restrict__ clusterWeight,
@@ -1477,10 +1490,10 @@ __global__ __launch_bounds__(256) void k4_sweep_hi_p1(
         const i64 ccSize = cci.size;
         double c0 = 0.0, selfLoop = 0.0;
         for (int k = lane; k < deg; k += 64) {
-            const i64 tail = tails[e0 + k]; // global == local at p==1
-            const double w = 1.0;           // hub path is unit-only
-            if (tail == v) selfLoop += w;
-            const i64 tcomm = (i64)currComm[sigma_inv[tail]];
+            const i64 ti = tidx32[e0 + k]; // pre-translated internal index
+            const double w = 1.0;          // hub path is unit-only
+            if (ti == i) selfLoop += w;
+            const i64 tcomm = (i64)currComm[ti];
             if (tcomm == (i64)cc) { c0 += w; continue; }
             i64 pos = (i64)(((uint64_t)tcomm * 0x9E3779B97F4A7C15ull) >> 32) &
                       (cap - 1);
@@ -1705,6 +1718,7 @@ struct mv_engine {
 
     // high-degree (wave-per-vertex) path: first nhi sorted positions
     i64 nhi = 0;
+    int *d_tidx32 = nullptr;   // lne pre-translated tails (hub kernels)
     i64 *d_hash_off = nullptr; // nhi+1 slot offsets (per-vertex caps are powers of two)
     i64 *d_hkeys = nullptr;
     double *d_hacc = nullptr;
@@ -1855,6 +1869,7 @@ static void free_graph_state(mv_engine *e) {
                      (void **)&e->d_spill_k, (void **)&e->d_spill_a,
                      (void **)&e->d_spill_off, (void **)&e->d_hash_off,
                      (void **)&e->d_hkeys, (void **)&e->d_hacc,
+                     (void **)&e->d_tidx32,
                      (void **)&e->d_last_sent, (void **)&e->d_chg_idx,
                      (void **)&e->d_chg_lab, (void **)&e->d_chg_cnt,
                      (void **)&e->d_cntmat, (void **)&e->d_soff,
@@ -2158,6 +2173,12 @@ static void build_sell(mv_engine *e) {
         while (nhi < lnv && sdeg[nhi] > HI_THRESH) nhi++;
         e->nhi = e->skewed ? nhi : 0;
         if (e->nhi > 0) {
+            if (!e->d_tidx32)
+                HIP_CHECK(hipMalloc(&e->d_tidx32,
+                                    4 * std::max<i64>(e->lne, 1)));
+            k_translate_tails<<<grid_for(e->lne), 256, 0, st>>>(
+                e->lne, e->d_tails, e->base, e->bound, e->d_sigma_inv,
+                e->d_ghosts, e->nghost, lnv, e->d_tidx32);
             std::vector<i64> hoff(e->nhi + 1);
             i64 acc = 0;
             for (i64 t = 0; t < e->nhi; t++) {
@@ -2894,16 +2915,16 @@ double mv_engine_run(mv_engine *e, double lower, double thresh,
                 k4_sweep_hi_p1<<<grid_for(e->nhi * 64, 256, 2048), 256, 0,
                                  st>>>(
                     e->nhi, lnv, e->d_perm, e->d_deg, e->d_sigma,
-                    e->d_sigma_inv, e->d_xadj, e->d_tails,
+                    e->d_xadj, e->d_tidx32,
                     (const unsigned *)d_curr, e->d_vdeg, e->d_cinfo,
                     e->d_cupd, constant, (unsigned *)d_target, e->d_cw,
                     e->d_hash_off, e->d_hkeys, e->d_hacc);
             else
             k4_sweep_hi_mr<<<grid_for(e->nhi * 64, 256, 2048), 256, 0,
                              st>>>(
-                e->nhi, lnv, e->base, e->bound, e->d_perm, e->d_deg,
-                e->d_sigma, e->d_sigma_inv, e->d_xadj, e->d_tails,
-                e->d_ghosts, e->nghost, e->d_vcurr, e->d_vghost, e->d_vdeg,
+                e->nhi, lnv, e->base, e->d_perm, e->d_deg,
+                e->d_sigma, e->d_xadj, e->d_tidx32,
+                e->d_vcurr, e->d_vghost, e->d_vdeg,
                 e->d_cinfo, e->d_cupd, e->d_rc_ids, e->d_rc_info,
                 e->d_rcu, constant, e->d_vtarget, e->d_cw, e->d_hash_off,
                 e->d_hkeys, e->d_hacc);
