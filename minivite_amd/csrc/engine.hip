@@ -2828,21 +2828,26 @@ double mv_engine_run(mv_engine *e, double lower, double thresh,
         // the first iterations avoids the spill path where it matters.
         // Tunables (perf only, results identical): MV_SLOTS, MV_SLOTS_FIRST,
         // MV_FIRST_ITERS.
-        static const int slots_rest = [] {
+        static const int slots_env = [] {
             const char *s = getenv("MV_SLOTS");
-            return s ? atoi(s) : 8;
+            return s ? atoi(s) : 0;
         }();
         static const int slots_first_env = [] {
             const char *s = getenv("MV_SLOTS_FIRST");
             return s ? atoi(s) : -1;
         }();
-        // slot mode's 12 B/lane LDS affords 12 slots (36 KiB/block) for
-        // the candidate-heavy first iterations: +2-3% whole-run, every
-        // size (same-box sweep). Handle mode keeps slots_rest (16 B/lane
-        // would drop occupancy harder).
+        // Degree-adaptive defaults: RGG-class graphs (deg ~10) measured
+        // best at 8 slots (12 first iterations at p==1); dense social
+        // shapes (deg ~70, ~25 persistent distinct candidates from long-
+        // range edges) at 24 (72 KiB/block, 2 blocks/CU) — +23% on the
+        // orkut_like workload, while 32+ falls off the occupancy cliff
+        // (experiments/RESULTS.md).
+        const bool dense = lne > 16 * lnv;
+        const int slots_rest = slots_env > 0 ? slots_env : (dense ? 24 : 8);
         const int slots_first =
-            slots_first_env > 0 ? slots_first_env
-                                : ((p == 1) ? 12 : slots_rest);
+            slots_first_env > 0
+                ? slots_first_env
+                : (dense ? 24 : ((p == 1) ? 12 : slots_rest));
         static const int first_iters = [] {
             const char *s = getenv("MV_FIRST_ITERS");
             return s ? atoi(s) : 2;
@@ -2877,6 +2882,8 @@ double mv_engine_run(mv_engine *e, double lower, double thresh,
             case 12: launch_sweep(std::integral_constant<int, 12>{}, unit_tag, s0, s1); break;
             case 16: launch_sweep(std::integral_constant<int, 16>{}, unit_tag, s0, s1); break;
             case 24: launch_sweep(std::integral_constant<int, 24>{}, unit_tag, s0, s1); break;
+            case 32: launch_sweep(std::integral_constant<int, 32>{}, unit_tag, s0, s1); break;
+            case 48: launch_sweep(std::integral_constant<int, 48>{}, unit_tag, s0, s1); break;
             default: launch_sweep(std::integral_constant<int, 8>{}, unit_tag, s0, s1); break;
             }
         };
