@@ -899,16 +899,20 @@ __global__ __launch_bounds__(256) void k4_sweep_hi_mr(
     }
 }
 
-// ---- K4 hub path for WEIGHTED skewed graphs (per-lane + hash) ----
-// The wave-per-vertex hub kernel needs order-free sums (unit weights); -w
-// hubs instead keep one LANE per vertex — accumulation stays in edge
-// order, bit-exact (dspl.hpp:240-271) — but the spill becomes an
-// open-addressed per-VERTEX hash (d_hash_off regions), turning the linear
-// spill's O(deg^2) probing into O(deg) expected. Works for p==1 too: the
-// u32 slot array IS the view there (base 0, no ghosts), so one kernel
-// serves both.
-__global__ __launch_bounds__(256) void k4_sweep_hubw(
-    i64 nhi, i64 lnv, i64 base, const unsigned *__restrict__ perm,
+// ---- K4 lane-hash path (skewed graphs, mid/high-degree band) ----
+// One LANE per vertex — accumulation stays in edge order, bit-exact for
+// -w (dspl.hpp:240-271) — with the clmap as an open-addressed per-VERTEX
+// hash (d_hash_off regions) PLUS a compact candidate list (d_lh_list):
+// O(1) expected insert/lookup instead of the LDS path's O(ncand) linear
+// probe per edge, and the argmax scans exactly ncand entries instead of
+// the whole hash capacity. This is the aggregation structure for
+// vertices whose persistent distinct-candidate count exceeds the LDS
+// slots (long-range-edge-heavy social graphs). Works for p==1 too: the
+// u32 slot array IS the view there (base 0, no ghosts).
+template <bool UNIT>
+__global__ __launch_bounds__(256) void k4_sweep_lh(
+    i64 s_begin, i64 s_end, i64 lnv, i64 base,
+    const unsigned *__restrict__ perm,
     const unsigned *__restrict__ deg_int, const i64 *__restrict__ chunk_off,
     const int *__restrict__ sell_tidx, const double *__restrict__ sell_w,
     const unsigned *__restrict__ vcurr, const unsigned *__restrict__ vghost,
@@ -918,19 +922,20 @@ __global__ __launch_bounds__(256) void k4_sweep_hubw(
     Info16 *__restrict__ rcu, double constant,
     unsigned *__restrict__ vtarget, double *__restrict__ clusterWeight,
     const i64 *__restrict__ hash_off, i64 *__restrict__ hkeys,
-    double *__restrict__ hacc) {
+    double *__restrict__ hacc, unsigned *__restrict__ lh_list) {
     const i64 gthread = blockIdx.x * (i64)blockDim.x + threadIdx.x;
     const i64 stride = (i64)gridDim.x * blockDim.x;
     auto label_of = [&](unsigned v) -> i64 {
         return (v < lnv) ? base + (i64)sigma[v] : rc_ids[clamp0(v - lnv)];
     };
-    for (i64 s = gthread; s < nhi; s += stride) {
+    for (i64 s = s_begin + gthread; s < s_end; s += stride) {
         const i64 i = perm[s];
         const int deg = (int)deg_int[i];
         const i64 ebase = chunk_off[s >> 6] + (s & 63);
         const unsigned cc = vcurr[i];
         const i64 hoff = hash_off[s];
         const i64 cap = hash_off[s + 1] - hoff; // power of two
+        unsigned *mylist = lh_list + (hoff >> 1); // cap/2 >= deg entries
         double ccDeg;
         i64 ccSize;
         if (cc < lnv) {
@@ -954,14 +959,14 @@ __global__ __launch_bounds__(256) void k4_sweep_hubw(
             for (int j = 0; j < CH; j++) {
                 const i64 slot = (j < m) ? ebase + (i64)(k0 + j) * 64 : ebase;
                 tb[j] = sell_tidx[slot];
-                wb[j] = sell_w[slot];
+                if (!UNIT) wb[j] = sell_w[slot];
             }
 #pragma unroll
             for (int j = 0; j < CH; j++)
                 cb[j] = (tb[j] < lnv) ? vcurr[tb[j]]
                                       : vghost[clamp0(tb[j] - lnv)];
             for (int j = 0; j < m; j++) {
-                const double w = wb[j];
+                const double w = UNIT ? 1.0 : wb[j];
                 if (tb[j] == i) selfLoop += w; // dspl.hpp:247-248
                 const unsigned tcomm = cb[j];
                 if (tcomm == cc) { c0 += w; continue; }
@@ -976,7 +981,7 @@ __global__ __launch_bounds__(256) void k4_sweep_hubw(
                     if (key == -1) {
                         hkeys[hoff + pos] = tcomm;
                         hacc[hoff + pos] = w;
-                        ncand++;
+                        mylist[ncand++] = (unsigned)pos;
                         break;
                     }
                     pos = (pos + 1) & (cap - 1);
@@ -985,8 +990,8 @@ __global__ __launch_bounds__(256) void k4_sweep_hubw(
         }
         clusterWeight[i] = c0; // dspl.hpp:318
 
-        // distGetMaxIndex over the hash (dspl.hpp:174-228). The scan order
-        // is hash-slot order, which is fine: the argmax with the lazy
+        // distGetMaxIndex over the compact list (dspl.hpp:174-228); the
+        // scan order is insertion order — fine: the argmax with the lazy
         // label tie-break is a total order, order-independent.
         const double vdeg = vDegree[i];
         const double eix = c0 - selfLoop;
@@ -994,13 +999,10 @@ __global__ __launch_bounds__(256) void k4_sweep_hubw(
         double maxGain = 0.0;
         unsigned maxIndex = cc;
         i64 maxSize = ccSize;
-        int seen = 0;
-        for (i64 t = 0; t < cap && seen < ncand; t++) {
-            const i64 yk = hkeys[hoff + t];
-            if (yk == -1) continue;
-            seen++;
-            const unsigned y = (unsigned)yk;
-            const double eiy = hacc[hoff + t];
+        for (int t = 0; t < ncand; t++) {
+            const i64 pos = mylist[t];
+            const unsigned y = (unsigned)hkeys[hoff + pos];
+            const double eiy = hacc[hoff + pos];
             double ay;
             i64 ysz;
             if (y < lnv) {
@@ -1716,13 +1718,19 @@ struct mv_engine {
     int skewed = 0;
     int sweep_grid = 0;
 
-    // high-degree (wave-per-vertex) path: first nhi sorted positions
+    // hash-aggregation bands over the degree-sorted positions:
+    // [0, nhi) = wave-per-vertex hubs (unit only), [nhi, nlh) = lane-hash
+    // band, [nlh, lnv) = LDS slots + linear spill
     i64 nhi = 0;
+    i64 nlh = 0;
+    unsigned *d_lh_list = nullptr; // compact candidate lists (hash/2)
     int *d_tidx32 = nullptr;   // lne pre-translated tails (hub kernels)
-    i64 *d_hash_off = nullptr; // nhi+1 slot offsets (per-vertex caps are powers of two)
+    i64 *d_hash_off = nullptr; // nlh+1 slot offsets (per-vertex caps are powers of two)
     i64 *d_hkeys = nullptr;
     double *d_hacc = nullptr;
     i64 hash_total = 0;
+    i64 hash_hub_elems = 0; // prefix the wave kernel atomics into (needs
+                            // zeroed accumulators each iteration)
 
     // trace
     i64 *trace_target = nullptr;
@@ -1869,7 +1877,7 @@ static void free_graph_state(mv_engine *e) {
                      (void **)&e->d_spill_k, (void **)&e->d_spill_a,
                      (void **)&e->d_spill_off, (void **)&e->d_hash_off,
                      (void **)&e->d_hkeys, (void **)&e->d_hacc,
-                     (void **)&e->d_tidx32,
+                     (void **)&e->d_tidx32, (void **)&e->d_lh_list,
                      (void **)&e->d_last_sent, (void **)&e->d_chg_idx,
                      (void **)&e->d_chg_lab, (void **)&e->d_chg_cnt,
                      (void **)&e->d_cntmat, (void **)&e->d_soff,
@@ -1902,6 +1910,7 @@ static void free_graph_state(mv_engine *e) {
     e->nghost = 0;
     e->ssz = 0;
     e->nhi = 0;
+    e->nlh = 0;
     e->cub_tmp_bytes = 0;
 }
 
@@ -2141,6 +2150,7 @@ static void build_sell(mv_engine *e) {
     k_degrees<<<grid_for(lnv), 256, 0, st>>>(lnv, e->d_sigma, e->d_xadj,
                                              e->d_deg);
     e->nhi = 0; // set by the skewed branch below when applicable
+    e->nlh = 0;
     if (e->has_hint && !e->skewed) {
         k_iota32<<<grid_for(lnv), 256, 0, st>>>(lnv, e->d_perm);
     } else {
@@ -2169,36 +2179,48 @@ static void build_sell(mv_engine *e) {
             const char *s = getenv("MV_HI_THRESH");
             return s ? (unsigned)atoi(s) : 256u;
         }();
+        // MV_LH_THRESH: degree bound above which a vertex takes the
+        // per-lane hash band (below: LDS slots + linear spill)
+        static const unsigned LH_THRESH = [] {
+            const char *s = getenv("MV_LH_THRESH");
+            return s ? (unsigned)atoi(s) : 24u;
+        }();
         i64 nhi = 0;
         while (nhi < lnv && sdeg[nhi] > HI_THRESH) nhi++;
-        e->nhi = e->skewed ? nhi : 0;
-        if (e->nhi > 0) {
+        i64 nlh = 0;
+        while (nlh < lnv && sdeg[nlh] > LH_THRESH) nlh++;
+        e->nhi = (e->skewed && e->unit_weights) ? nhi : 0;
+        e->nlh = e->skewed ? std::max<i64>(nlh, e->nhi) : 0;
+        if (e->nlh > 0) {
             if (!e->d_tidx32)
                 HIP_CHECK(hipMalloc(&e->d_tidx32,
                                     4 * std::max<i64>(e->lne, 1)));
             k_translate_tails<<<grid_for(e->lne), 256, 0, st>>>(
                 e->lne, e->d_tails, e->base, e->bound, e->d_sigma_inv,
                 e->d_ghosts, e->nghost, lnv, e->d_tidx32);
-            std::vector<i64> hoff(e->nhi + 1);
+            std::vector<i64> hoff(e->nlh + 1);
             i64 acc = 0;
-            for (i64 t = 0; t < e->nhi; t++) {
+            for (i64 t = 0; t < e->nlh; t++) {
                 hoff[t] = acc;
                 i64 cap = 64;
                 while (cap < 2 * (i64)sdeg[t]) cap <<= 1;
                 acc += cap;
             }
-            hoff[e->nhi] = acc;
-            if (!e->d_hash_off)
-                HIP_CHECK(hipMalloc(&e->d_hash_off, 8 * (e->nhi + 1)));
+            hoff[e->nlh] = acc;
+            e->hash_hub_elems = e->nhi > 0 ? hoff[e->nhi] : 0;
+            if (e->d_hash_off) HIP_CHECK(hipFree(e->d_hash_off));
+            HIP_CHECK(hipMalloc(&e->d_hash_off, 8 * (e->nlh + 1)));
             HIP_CHECK(hipMemcpyAsync(e->d_hash_off, hoff.data(),
-                                     8 * (e->nhi + 1),
+                                     8 * (e->nlh + 1),
                                      hipMemcpyHostToDevice, st));
             if (acc > e->hash_total) {
                 if (e->d_hkeys) HIP_CHECK(hipFree(e->d_hkeys));
                 if (e->d_hacc) HIP_CHECK(hipFree(e->d_hacc));
+                if (e->d_lh_list) HIP_CHECK(hipFree(e->d_lh_list));
                 e->hash_total = acc;
                 HIP_CHECK(hipMalloc(&e->d_hkeys, 8 * acc));
                 HIP_CHECK(hipMalloc(&e->d_hacc, 8 * acc));
+                HIP_CHECK(hipMalloc(&e->d_lh_list, 4 * (acc / 2 + 1)));
             }
             HIP_CHECK(hipStreamSynchronize(st));
         }
@@ -2315,7 +2337,7 @@ static void build_sell(mv_engine *e) {
             i64 *d_need = nullptr;
             HIP_CHECK(hipMalloc(&d_need, 8 * nthreads));
             k_spill_need<<<grid_for(nthreads), 256, 0, st>>>(
-                nthreads, e->nhi, lnv, e->d_perm, e->d_deg, 4, d_need);
+                nthreads, e->nlh, lnv, e->d_perm, e->d_deg, 4, d_need);
             std::vector<i64> need(nthreads), off(nthreads);
             HIP_CHECK(hipMemcpyAsync(need.data(), d_need, 8 * nthreads,
                                      hipMemcpyDeviceToHost, st));
@@ -2906,19 +2928,34 @@ double mv_engine_run(mv_engine *e, double lower, double thresh,
                     e->d_ghosts, e->d_vdeg, e->d_sigma, e->d_cupd,
                     e->d_rc_info, e->d_rcu, constant, e->d_vtarget, e->d_cw);
         };
-        if (e->nhi > 0) { // hub path (skewed graphs)
-            HIP_CHECK(hipMemsetAsync(e->d_hkeys, 0xFF, 8 * e->hash_total, st));
-            HIP_CHECK(hipMemsetAsync(e->d_hacc, 0, 8 * e->hash_total, st));
-            if (!e->unit_weights) // -w: per-lane serial hash (edge order)
-                k4_sweep_hubw<<<grid_for(e->nhi), 256, 0, st>>>(
-                    e->nhi, lnv, e->base, e->d_perm, e->d_deg,
-                    e->d_chunk_off, e->d_sell_tidx, e->d_sell_w,
-                    p == 1 ? (const unsigned *)d_curr : e->d_vcurr,
-                    e->d_vghost, e->d_vdeg, e->d_sigma, e->d_cinfo,
-                    e->d_cupd, e->d_rc_ids, e->d_rc_info, e->d_rcu,
-                    constant, p == 1 ? (unsigned *)d_target : e->d_vtarget,
-                    e->d_cw, e->d_hash_off, e->d_hkeys, e->d_hacc);
-            else if (p == 1)
+        if (e->nlh > 0) { // hash-aggregation bands (skewed graphs)
+            HIP_CHECK(hipMemsetAsync(e->d_hkeys, 0xFF, 8 * e->hash_total,
+                                     st));
+            if (e->hash_hub_elems) // the wave kernel atomic-adds into its
+                                   // prefix; the lane band writes on insert
+                HIP_CHECK(hipMemsetAsync(e->d_hacc, 0,
+                                         8 * e->hash_hub_elems, st));
+            if (e->nlh > e->nhi) { // per-lane hash band [nhi, nlh)
+                auto launch_lh = [&](auto unit_tag) {
+                    k4_sweep_lh<decltype(unit_tag)::value>
+                        <<<grid_for(e->nlh - e->nhi), 256, 0, st>>>(
+                            e->nhi, e->nlh, lnv, e->base, e->d_perm,
+                            e->d_deg, e->d_chunk_off, e->d_sell_tidx,
+                            e->d_sell_w,
+                            p == 1 ? (const unsigned *)d_curr : e->d_vcurr,
+                            e->d_vghost, e->d_vdeg, e->d_sigma, e->d_cinfo,
+                            e->d_cupd, e->d_rc_ids, e->d_rc_info, e->d_rcu,
+                            constant,
+                            p == 1 ? (unsigned *)d_target : e->d_vtarget,
+                            e->d_cw, e->d_hash_off, e->d_hkeys, e->d_hacc,
+                            e->d_lh_list);
+                };
+                if (e->unit_weights)
+                    launch_lh(std::integral_constant<bool, true>{});
+                else
+                    launch_lh(std::integral_constant<bool, false>{});
+            }
+            if (e->nhi > 0 && p == 1)
                 k4_sweep_hi_p1<<<grid_for(e->nhi * 64, 256, 2048), 256, 0,
                                  st>>>(
                     e->nhi, lnv, e->d_perm, e->d_deg, e->d_sigma,
@@ -2926,7 +2963,7 @@ double mv_engine_run(mv_engine *e, double lower, double thresh,
                     (const unsigned *)d_curr, e->d_vdeg, e->d_cinfo,
                     e->d_cupd, constant, (unsigned *)d_target, e->d_cw,
                     e->d_hash_off, e->d_hkeys, e->d_hacc);
-            else
+            else if (e->nhi > 0)
             k4_sweep_hi_mr<<<grid_for(e->nhi * 64, 256, 2048), 256, 0,
                              st>>>(
                 e->nhi, lnv, e->base, e->d_perm, e->d_deg,
@@ -2978,7 +3015,7 @@ double mv_engine_run(mv_engine *e, double lower, double thresh,
             writeback(e->nexp, lnv);
             HIP_CHECK(hipEventRecord(e->ev_p2, st));
         } else {
-            sweep_range(e->nhi, lnv);
+            sweep_range(e->nlh, lnv);
             writeback(0, lnv);
         }
         HIP_CHECK(hipEventRecord(ev1, st));

@@ -205,7 +205,7 @@ def test_skewed_degree_graph_parity():
 
 def test_skewed_weighted_graph_parity():
     """Weighted power-law graph (hub degree ~10k): exercises the per-lane
-    hash-spill hub path (k4_sweep_hubw — bit-exact edge-order -w sums with
+    hash-spill hub path (the lane-hash kernel — bit-exact edge-order -w sums with
     O(deg) probing); engine vs oracle on the identical from_csr input."""
     import numpy as np
     from minivite_amd import Graph, Engine
