@@ -2180,10 +2180,15 @@ static void build_sell(mv_engine *e) {
             return s ? (unsigned)atoi(s) : 256u;
         }();
         // MV_LH_THRESH: degree bound above which a vertex takes the
-        // per-lane hash band (below: LDS slots + linear spill)
+        // per-lane hash band (below: LDS slots + linear spill). Default ==
+        // HI_THRESH: the band only covers WEIGHTED hubs (where it replaces
+        // the old cap-scanning kernel); extending it to the unit 24-256
+        // band was measured SLOWER than 24 LDS slots + linear spill on the
+        // dense social shape (11.3 -> 8.3 G, experiments/RESULTS.md) — L2
+        // hash probes lose to LDS at ~25 candidates.
         static const unsigned LH_THRESH = [] {
             const char *s = getenv("MV_LH_THRESH");
-            return s ? (unsigned)atoi(s) : 24u;
+            return s ? (unsigned)atoi(s) : 256u;
         }();
         i64 nhi = 0;
         while (nhi < lnv && sdeg[nhi] > HI_THRESH) nhi++;
