@@ -383,15 +383,16 @@ def test_loopback_p3_nonpow2():
     _loopback_vs_oracle(nv, world, csrs, parts, trace_cap=256)
 
 
-@pytest.mark.parametrize("trial", [0, 1, 2])
+@pytest.mark.parametrize("trial", [0, 1, 2, 3, 4])
 def test_loopback_fuzz(trial):
     """Randomized small graphs (self-loops, parallel edges, isolated
-    vertices, mixed weights) x rank counts {2,3,4}: the multi-rank engine
-    must match the oracle bit-for-bit on arbitrary partitioned inputs."""
+    vertices, mixed weights) x rank counts {2,3,4,6,8}: the multi-rank
+    engine must match the oracle bit-for-bit on arbitrary partitioned
+    inputs."""
     rng = np.random.default_rng(1000 + trial)
     nv = int(rng.integers(2000, 8000))
-    world = [2, 3, 4][trial]
-    unit = trial != 1
+    world = [2, 3, 4, 6, 8][trial]
+    unit = trial not in (1, 3)
     m = nv * int(rng.integers(3, 9))
     u = rng.integers(0, nv, m)
     v = rng.integers(0, nv, m)
