@@ -1,14 +1,16 @@
 """Multi-rank HIP path on ONE GPU via the loopback transport.
 
 RCCL refuses two ranks on one device, so the p>1 engine code — the
-handle-mode K4 sweep (engine.hip k4_sweep / k4_sweep_iter1), k8 community
-gathers, label<->handle conversion, candidate filter + sort/unique, K9
-replies, per-sender delta application, and the whole per-iteration exchange
-protocol (dspl.hpp:497-1103 equivalents) — is executed here with the
-collectives replaced by device-to-device copies + host barriers
-(mv_lb_session). Every kernel and every byte layout is the production p>1
-path; only ncclSend/Recv itself is substituted. Parity: bit-exact vs the
-reference-pinned oracle fixtures (tests/golden/pins.json).
+view-mode K4 sweeps (engine.hip k4_sweep_mr / k4_sweep_iter1_mr / the hub
+kernels), k8 community gathers, view builds, candidate filter +
+sort/unique, K9 replies, delta compaction, per-sender delta application,
+and the whole deep-pipelined per-iteration exchange protocol
+(dspl.hpp:497-1103 equivalents) — is executed here with the collectives
+replaced by device-to-device copies + host barriers (mv_lb_session).
+Every kernel and every byte layout is the production p>1 path; only
+ncclSend/Recv itself is substituted. Parity: bit-exact vs the
+reference-pinned oracle fixtures (tests/golden/pins.json) or the oracle
+on identical partitioned inputs.
 """
 import json
 import os
