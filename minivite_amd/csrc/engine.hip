@@ -1911,6 +1911,7 @@ static void free_graph_state(mv_engine *e) {
     e->ssz = 0;
     e->nhi = 0;
     e->nlh = 0;
+    e->hash_hub_elems = 0;
     e->cub_tmp_bytes = 0;
 }
 
