@@ -101,7 +101,7 @@ struct mv_graph {
 static void build_bfs_hint(mv_graph *g) {
     if (getenv("MV_NO_BFS_HINT")) return;
     const int64_t lnv = (int64_t)g->xadj.size() - 1;
-    if (lnv <= 1) return;
+    if (lnv <= 1 || lnv >= (1ll << 31)) return; // int32 labels below
     const int64_t base = g->parts[g->rank], bound = g->parts[g->rank + 1];
     // 2 rounds of async label propagation over the LOCAL subgraph
     // (deterministic sequential order; timestamped counters keep it
